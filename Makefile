@@ -1,0 +1,65 @@
+# Makefile — builds the MI355X-native core: the _hpk Python extension and the
+# standalone miniapp binaries. Everything cross-compiles for gfx950 with hipcc
+# (no GPU needed to build).
+#
+#   make -j        # extension + binaries
+#   make ext       # just the Python extension
+#   make bins      # just the standalone binaries
+
+HIPCC      ?= /opt/rocm/bin/hipcc
+GPU_ARCH   ?= gfx950
+PYTHON     ?= python3
+
+NATIVE     := hpc_patterns_amd/native
+BUILD      := build
+BIN        := bin
+
+PY_INC     := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PYBIND_INC := $(shell $(PYTHON) -c "import pybind11; print(pybind11.get_include())")
+EXT_SUFFIX := .so
+
+CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(GPU_ARCH) -I$(NATIVE) -Wall
+LDFLAGS    := -L/opt/rocm/lib -lrocm_smi64
+
+LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip
+LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
+
+EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)
+BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p
+
+.PHONY: all ext bins clean
+all: ext bins
+ext: $(EXT_SO)
+bins: $(BINARIES)
+
+$(BUILD)/%.o: $(NATIVE)/%.hip $(NATIVE)/include/hpk.h | $(BUILD)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/ext.o: $(NATIVE)/ext.cpp $(NATIVE)/include/hpk.h | $(BUILD)
+	$(HIPCC) $(CXXFLAGS) -I$(PY_INC) -I$(PYBIND_INC) -c $< -o $@
+
+$(EXT_SO): $(BUILD)/ext.o $(LIB_OBJS)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) -shared $^ -o $@ $(LDFLAGS)
+
+$(BUILD)/%_main.o: cpp/%_main.cpp $(NATIVE)/include/hpk.h | $(BUILD)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BIN)/hpk_conc: $(BUILD)/conc_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS)
+
+$(BIN)/hpk_topology: $(BUILD)/topology_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS)
+
+$(BIN)/hpk_allreduce: $(BUILD)/allreduce_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
+
+$(BIN)/hpk_p2p: $(BUILD)/p2p_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
+
+$(BUILD):
+	mkdir -p $(BUILD)
+$(BIN):
+	mkdir -p $(BIN)
+
+clean:
+	rm -rf $(BUILD) $(BIN) $(EXT_SO)

@@ -1,0 +1,312 @@
+// allreduce_main.cpp — hpk_allreduce: device-buffer all-reduce miniapp.
+//
+// MI355X-native re-design of the reference GPU-aware-MPI ring-allreduce
+// miniapps (reference aurora.mpich.miniapps/src/allreduce/mpi-sycl/
+// allreduce-mpi-sycl.cpp:88-215 and the two OMP variants). The reference's
+// three comparisons (hand ring over blocking MPI_Send/Recv vs native
+// MPI_Allreduce) become, on MI355X:
+//   --algo ring     hand ring over RCCL ncclSend/ncclRecv pt2pt + the
+//                   hand-written HIP Accumulate kernel (K3) between steps —
+//                   the reference's SendRecvRing+Accumulate pattern
+//                   (allreduce-mpi-sycl.cpp:44-59,173-182)
+//   --algo pipeline chunked ring: send/recv and Accumulate overlapped on two
+//                   hipStreams (the tuned variant the reference leaves as an
+//                   exercise — SURVEY.md §7.4)
+//   --algo rccl     native ncclAllReduce over xGMI (the reference's
+//                   MPI_Allreduce path, allreduce-mpi-sycl.cpp:62-67)
+//
+// Launcher model: no MPI exists on this stack; the binary self-launches
+// one process per GPU (fork), sharing the ncclUniqueId through pre-fork
+// memory — the MI355X-native "mpirun -np N" for a single node.
+//
+// Verification is the reference's analytic oracle: VA=rank everywhere, so
+// after all-reduce every element must equal size*(size-1)/2
+// (allreduce-mpi-sycl.cpp:192-204) — checked with an exact device-side
+// double reduction instead of an O(N) host scan.
+//
+// CLI (reference getopt surface, allreduce-mpi-sycl.cpp:106-131, extended):
+//   -p <P>     2^P elements (default 25 -> 128 MiB float)
+//   -D|-H|-S   allocator: hipMalloc | hipHostMalloc | hipMallocManaged
+//   -a         native ncclAllReduce (same as --algo rccl)
+//   -n <N>     number of ranks (default: visible device count)
+//   -i <iters> timed iterations (default 10, min-time reported)
+//   -c <K>     pipeline chunk count (default 8; pipeline algo only)
+//   --algo ring|pipeline|rccl
+
+#include "../hpc_patterns_amd/native/include/hpk.h"
+
+#include <rccl/rccl.h>
+#include <sys/mman.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cmath>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+namespace {
+
+void check_nccl(ncclResult_t r, const char* what) {
+  if (r != ncclSuccess) {
+    std::fprintf(stderr, "RCCL error in %s: %s\n", what,
+                 ncclGetErrorString(r));
+    std::exit(1);
+  }
+}
+
+double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+struct Config {
+  int p = 25;
+  char alloc = 'D';
+  std::string algo = "ring";
+  int nranks = -1;
+  int iters = 10;
+  int chunks = 8;
+};
+
+void* alloc_buf(char kind, size_t bytes) {
+  void* p = nullptr;
+  switch (kind) {
+    case 'D': hpk::check_hip(hipMalloc(&p, bytes), "hipMalloc"); break;
+    case 'H':
+      hpk::check_hip(hipHostMalloc(&p, bytes, hipHostMallocDefault),
+                     "hipHostMalloc");
+      break;
+    case 'S':
+      hpk::check_hip(hipMallocManaged(&p, bytes, hipMemAttachGlobal),
+                     "hipMallocManaged");
+      break;
+    default: std::abort();
+  }
+  return p;
+}
+
+// Hand ring all-reduce, the reference SendRecvRing pattern: (size-1) steps of
+// [exchange full buffer with ring neighbours] + [VC += recv]. Buffers VA
+// (send payload, swapped with VB each step), VB (recv), VC (accumulator).
+double run_ring(ncclComm_t comm, hipStream_t stream, float* va, float* vb,
+                float* vc, size_t n, int rank, int size) {
+  int right = (rank + 1) % size;
+  int left = (rank - 1 + size) % size;
+  double t0 = now_s();
+  hpk::launch_acc_f32(vc, va, n, stream); // VC += own VA (VC starts at 0)
+  for (int step = 0; step < size - 1; ++step) {
+    // RCCL pt2pt: group makes the send+recv concurrent (no odd/even ordering
+    // dance needed — that deadlock-avoidance trick is an MPI-blocking-call
+    // artifact, reference allreduce-mpi-sycl.cpp:50-58).
+    check_nccl(ncclGroupStart(), "group start");
+    check_nccl(ncclSend(va, n, ncclFloat, right, comm, stream), "send");
+    check_nccl(ncclRecv(vb, n, ncclFloat, left, comm, stream), "recv");
+    check_nccl(ncclGroupEnd(), "group end");
+    hpk::launch_acc_f32(vc, vb, n, stream);
+    std::swap(va, vb);
+  }
+  hpk::check_hip(hipStreamSynchronize(stream), "ring sync");
+  return now_s() - t0;
+}
+
+// Chunked pipelined ring: split the buffer into K chunks; while chunk c is
+// being accumulated on the compute stream, chunk c+1 is already in flight on
+// the comm stream. Overlaps xGMI transfer with the HIP Accumulate kernel.
+double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
+                    hipStream_t comp_stream, float* va, float* vb, float* vc,
+                    size_t n, int rank, int size, int chunks) {
+  int right = (rank + 1) % size;
+  int left = (rank - 1 + size) % size;
+  size_t chunk = (n + chunks - 1) / chunks;
+  std::vector<hipEvent_t> done((size_t)chunks);
+  for (auto& e : done)
+    hpk::check_hip(hipEventCreateWithFlags(&e, hipEventDisableTiming), "ev");
+
+  double t0 = now_s();
+  hpk::launch_acc_f32(vc, va, n, comp_stream);
+  for (int step = 0; step < size - 1; ++step) {
+    // launch all chunk exchanges; accumulate each chunk as soon as it lands
+    for (int c = 0; c < chunks; ++c) {
+      size_t off = (size_t)c * chunk;
+      if (off >= n) break;
+      size_t len = std::min(chunk, n - off);
+      check_nccl(ncclGroupStart(), "group start");
+      check_nccl(ncclSend(va + off, len, ncclFloat, right, comm, comm_stream),
+                 "send");
+      check_nccl(ncclRecv(vb + off, len, ncclFloat, left, comm, comm_stream),
+                 "recv");
+      check_nccl(ncclGroupEnd(), "group end");
+      hpk::check_hip(hipEventRecord(done[c], comm_stream), "record");
+      hpk::check_hip(hipStreamWaitEvent(comp_stream, done[c], 0), "wait");
+      hpk::launch_acc_f32(vc + off, vb + off, len, comp_stream);
+    }
+    // comm of next step must not overwrite vb before accumulate read it:
+    // swap uses distinct buffers, but step s+1 recv into (old) va after
+    // compute consumed it -> order via event from comp_stream.
+    hipEvent_t barrier_ev = done[0];
+    hpk::check_hip(hipEventRecord(barrier_ev, comp_stream), "step ev");
+    hpk::check_hip(hipStreamWaitEvent(comm_stream, barrier_ev, 0), "step wait");
+    std::swap(va, vb);
+  }
+  hpk::check_hip(hipStreamSynchronize(comp_stream), "pipeline sync");
+  hpk::check_hip(hipStreamSynchronize(comm_stream), "pipeline sync comm");
+  double dt = now_s() - t0;
+  for (auto& e : done) (void)hipEventDestroy(e);
+  return dt;
+}
+
+int worker(int rank, int size, int ndev, const ncclUniqueId& id,
+           const Config& cfg, double* shared_out) {
+  int dev = rank % ndev;
+  hpk::check_hip(hipSetDevice(dev), "hipSetDevice");
+  ncclComm_t comm;
+  check_nccl(ncclCommInitRank(&comm, size, const_cast<ncclUniqueId&>(id), rank),
+             "ncclCommInitRank");
+
+  size_t n = 1ull << cfg.p;
+  size_t bytes = n * sizeof(float);
+  float* va = (float*)alloc_buf(cfg.alloc, bytes);
+  float* vb = (float*)alloc_buf(cfg.alloc, bytes);
+  float* vc = (float*)alloc_buf(cfg.alloc, bytes);
+
+  hipStream_t stream, comp_stream;
+  hpk::check_hip(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking), "s");
+  hpk::check_hip(hipStreamCreateWithFlags(&comp_stream, hipStreamNonBlocking),
+                 "s2");
+
+  double best = 1e30;
+  for (int it = 0; it < cfg.iters; ++it) {
+    // (re)initialize: VA = rank, VB = -1, VC = 0 (reference Initialize)
+    hpk::launch_fill_f32(va, (float)rank, n, stream);
+    hpk::launch_fill_f32(vb, -1.f, n, stream);
+    hpk::launch_fill_f32(vc, 0.f, n, stream);
+    hpk::check_hip(hipStreamSynchronize(stream), "init sync");
+
+    double dt;
+    if (cfg.algo == "rccl") {
+      double t0 = now_s();
+      check_nccl(ncclAllReduce(va, vc, n, ncclFloat, ncclSum, comm, stream),
+                 "ncclAllReduce");
+      hpk::check_hip(hipStreamSynchronize(stream), "allreduce sync");
+      dt = now_s() - t0;
+    } else if (cfg.algo == "pipeline") {
+      dt = run_pipeline(comm, stream, comp_stream, va, vb, vc, n, rank, size,
+                        cfg.chunks);
+    } else {
+      dt = run_ring(comm, stream, va, vb, vc, n, rank, size);
+    }
+    best = std::min(best, dt);
+  }
+
+  // max over ranks (reference MPI_Allreduce MAX of wall time)
+  double* d_time = nullptr;
+  hpk::check_hip(hipMalloc(&d_time, sizeof(double)), "time buf");
+  hpk::check_hip(hipMemcpy(d_time, &best, sizeof(double), hipMemcpyHostToDevice),
+                 "time h2d");
+  check_nccl(ncclAllReduce(d_time, d_time, 1, ncclDouble, ncclMax, comm, stream),
+             "time max");
+  hpk::check_hip(hipStreamSynchronize(stream), "time sync");
+  double max_time;
+  hpk::check_hip(hipMemcpy(&max_time, d_time, sizeof(double),
+                           hipMemcpyDeviceToHost),
+                 "time d2h");
+
+  // analytic verification: every element == size*(size-1)/2
+  double expected = (double)n * ((double)size * (size - 1) / 2.0);
+  double got = hpk::reduce_sum_f32(vc, n, stream);
+  bool pass = std::abs(got - expected) < 1e-6 * std::max(1.0, expected);
+  std::printf("%s rank %d (sum %.1f, expected %.1f)\n",
+              pass ? "Passed" : "FAILED", rank, got, expected);
+
+  if (rank == 0) {
+    double gb = (double)bytes / 1e9;
+    // bus bandwidth convention: ring moves 2(size-1)/size * bytes per rank
+    double busbw =
+        size > 1 ? 2.0 * (size - 1) / size * gb / max_time : gb / max_time;
+    std::printf("# algo=%s ranks=%d elems=2^%d alloc=%c time=%.6fs "
+                "busbw=%.2f GB/s\n",
+                cfg.algo.c_str(), size, cfg.p, cfg.alloc, max_time, busbw);
+    if (shared_out) *shared_out = max_time;
+  }
+
+  ncclCommDestroy(comm);
+  (void)hipFree(d_time);
+  return pass ? 0 : 2;
+}
+
+} // namespace
+
+int main(int argc, char* argv[]) {
+  Config cfg;
+  for (int i = 1; i < argc; ++i) {
+    std::string s = argv[i];
+    auto next = [&]() -> const char* {
+      if (++i >= argc) { std::fprintf(stderr, "missing value\n"); std::exit(1); }
+      return argv[i];
+    };
+    if (s == "-p") cfg.p = std::atoi(next());
+    else if (s == "-D") cfg.alloc = 'D';
+    else if (s == "-H") cfg.alloc = 'H';
+    else if (s == "-S") cfg.alloc = 'S';
+    else if (s == "-a") cfg.algo = "rccl";
+    else if (s == "-n") cfg.nranks = std::atoi(next());
+    else if (s == "-i") cfg.iters = std::atoi(next());
+    else if (s == "-c") cfg.chunks = std::atoi(next());
+    else if (s == "--algo") cfg.algo = next();
+    else {
+      std::printf(
+          "Usage: %s [-p P] [-D|-H|-S] [-a] [-n ranks] [-i iters] [-c chunks] "
+          "[--algo ring|pipeline|rccl]\n", argv[0]);
+      return s == "-h" || s == "--help" ? 0 : 1;
+    }
+  }
+  if (cfg.algo != "ring" && cfg.algo != "pipeline" && cfg.algo != "rccl") {
+    std::fprintf(stderr, "unknown algo '%s'\n", cfg.algo.c_str());
+    return 1;
+  }
+
+  int ndev = hpk::device_count();
+  if (ndev == 0) {
+    std::fprintf(stderr, "no HIP devices\n");
+    return 1;
+  }
+  int size = cfg.nranks > 0 ? cfg.nranks : ndev;
+  if (size > ndev) {
+    // RCCL refuses two ranks on one device; clamp with a notice instead of
+    // oversubscribing (the reference oversubscribed tiles; xGMI pt2pt
+    // between co-located ranks would be a self-copy anyway).
+    std::fprintf(stderr, "# clamping ranks %d -> %d (one rank per GPU)\n",
+                 size, ndev);
+    size = ndev;
+  }
+
+  ncclUniqueId id;
+  check_nccl(ncclGetUniqueId(&id), "ncclGetUniqueId");
+
+  // shared page for rank0's timing result (read by the parent after waitpid)
+  double* shared_out =
+      (double*)mmap(nullptr, sizeof(double), PROT_READ | PROT_WRITE,
+                    MAP_SHARED | MAP_ANONYMOUS, -1, 0);
+
+  if (size == 1) return worker(0, 1, ndev, id, cfg, shared_out);
+
+  std::vector<pid_t> pids;
+  for (int r = 0; r < size; ++r) {
+    pid_t pid = fork();
+    if (pid == 0) std::exit(worker(r, size, ndev, id, cfg, shared_out));
+    pids.push_back(pid);
+  }
+  int rc = 0;
+  for (pid_t pid : pids) {
+    int st = 0;
+    waitpid(pid, &st, 0);
+    if (!WIFEXITED(st) || WEXITSTATUS(st) != 0) rc = 1;
+  }
+  return rc;
+}

@@ -1,0 +1,133 @@
+// hpk.h — C++ API of the MI355X-native HPC-patterns core library.
+//
+// Brand-new CDNA4/HIP design with the capabilities of the reference suite
+// (argonne-lcf/HPC-Patterns): hand-written gfx950 kernels (the reference's
+// SYCL parallel_for / omp-target loops, see reference concurency/bench.hpp:7-31,
+// allreduce-mpi-sycl.cpp:27-41), a multi-hipStream/hipGraph concurrency engine
+// (reference bench_sycl.cpp:19-144 / bench_omp.cpp), xGMI topology discovery
+// (reference p2p/topology.cpp, Level-Zero Sysman -> rocm_smi/HIP here), and
+// HIP-IPC one-sided transport (reference MPI_Win/MPI_Put path,
+// p2p/peer2pear.cpp:68-102).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstddef>
+#include <cstdint>
+#include <map>
+#include <string>
+#include <vector>
+
+namespace hpk {
+
+// ---------------------------------------------------------------------------
+// Error handling
+// ---------------------------------------------------------------------------
+void check_hip(hipError_t e, const char* what);
+
+// ---------------------------------------------------------------------------
+// Kernels (kernels.hip) — all launch asynchronously on `stream` unless noted.
+// ---------------------------------------------------------------------------
+
+// K1 "C" compute command: every work-item runs 64*tripcount dependent FMAs on
+// registers and stores one float (reference bench.hpp:23-31 MAD_64 payload).
+void launch_busy_wait(float* out, long tripcount, long globalsize,
+                      hipStream_t stream);
+
+// K1-MFMA variant: every wave runs `tripcount` chained
+// v_mfma_f32_16x16x32_bf16 ops — lights up the matrix cores so concurrency
+// profiles show MFMA utilisation. n_waves waves of 64 lanes.
+void launch_busy_wait_mfma(float* out, long tripcount, long n_waves,
+                           hipStream_t stream);
+
+// K2 shader-copy: vectorized 16B grid-stride copy (the "shader blit" sibling
+// of hipMemcpyAsync's SDMA path). Pointers may be in any HIP-visible space.
+void launch_copy_kernel(void* dst, const void* src, size_t nbytes,
+                        hipStream_t stream);
+
+// K4 fills (reference Initialize kernel, allreduce-mpi-sycl.cpp:34-41).
+void launch_fill_f32(float* dst, float value, size_t n, hipStream_t stream);
+// dst[i] = (float)i — device-side iota payload generator (reference
+// fill_randomly host shuffle, peer2pear.cpp:8-17; shuffling is done on host,
+// the plain iota fill runs on device).
+void launch_iota_f32(float* dst, size_t n, hipStream_t stream);
+
+// K3 accumulate: dst[i] += src[i] (reference Accumulate,
+// allreduce-mpi-sycl.cpp:27-31), vectorized float4 grid-stride.
+void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream);
+
+// Exact double-precision sum of n floats. Synchronizes `stream`.
+// Replaces the reference's O(N log N) host sort+sum checksum
+// (peer2pear.cpp:56-63) with an order-independent exact device reduction.
+double reduce_sum_f32(const float* src, size_t n, hipStream_t stream);
+
+// ---------------------------------------------------------------------------
+// Concurrency engine (conc.hip) — reference bench<T>() ABI, bench.hpp:37-40.
+// ---------------------------------------------------------------------------
+
+// Modes:
+//   serial       — one stream, synchronize after every command (baseline)
+//   in_order     — N hipStreams (default: one per command), round-robin
+//   host_threads — one std::thread + stream per command (reference
+//                  bench_omp.cpp host_threads mode)
+//   graph        — all commands as independent nodes of one hipGraph; the
+//                  HIP analog of a SYCL out-of-order queue
+//   out_of_order — alias of graph (HIP streams are strictly in-order; the
+//                  graph scheduler is the runtime-managed concurrency path)
+//   nowait       — alias of in_order (reference bench_omp.cpp nowait mode)
+extern const std::string allowed_modes;
+bool mode_is_allowed(const std::string& mode);
+
+struct ConcResult {
+  long total_us = 0;                  // min over repetitions
+  std::vector<long> per_cmd_us;       // serial mode: min per-command wall time
+  std::vector<double> per_cmd_dev_ms; // hipEvent device time (profiling only)
+};
+
+// commands: "C" or "A2B"/"AB" with A,B in {M,D,H,S} =
+// malloc/hipMalloc/hipHostMalloc/hipMallocManaged (reference
+// bench_sycl.cpp:55-72 letter DSL).
+// params keys: tripcount_C, globalsize_C, globalsize_<CMD> (floats).
+ConcResult conc_bench(const std::string& mode,
+                      const std::vector<std::string>& commands,
+                      const std::map<std::string, size_t>& params,
+                      bool enable_profiling, int n_queues, int n_repetitions,
+                      bool verbose, bool use_copy_kernel);
+
+// ---------------------------------------------------------------------------
+// Topology (topo.hip) — xGMI link discovery (reference p2p/topology.cpp).
+// ---------------------------------------------------------------------------
+
+struct LinkInfo {
+  int p2p_accessible = 0; // hipDeviceCanAccessPeer
+  int link_type = -1;     // hipExtGetLinkTypeAndHopCount type (2 == xGMI)
+  int hops = -1;
+  long min_bw_mbps = -1;  // rocm_smi min/max link bandwidth (if available)
+  long max_bw_mbps = -1;
+  long weight = -1;       // rocm_smi link weight (if available)
+};
+
+int device_count();
+// NxN matrix; [i][i] is zero-initialized LinkInfo.
+std::vector<std::vector<LinkInfo>> link_matrix();
+// Connected components under direct-P2P reachability (the reference's
+// "connectivity planes", topology.cpp:76-89). MI355X nodes are fully
+// connected, so this is usually one plane — the interesting data is the
+// per-pair link table above.
+std::vector<std::vector<int>> p2p_planes();
+
+// ---------------------------------------------------------------------------
+// IPC one-sided transport (ipc.hip) — reference MPI_Win/MPI_Put analog.
+// ---------------------------------------------------------------------------
+
+// 64-byte opaque handle for a hipMalloc'd region, exchangeable between
+// processes (dmabuf IPC; requires HSA_ENABLE_IPC_MODE_LEGACY=0).
+std::vector<uint8_t> ipc_get_handle(void* dptr);
+void* ipc_open_handle(const std::vector<uint8_t>& handle);
+void ipc_close_handle(void* dptr);
+
+void enable_peer_access(int peer_device);
+void memcpy_peer_async(void* dst, int dst_dev, const void* src, int src_dev,
+                       size_t nbytes, hipStream_t stream);
+
+} // namespace hpk

@@ -1,0 +1,289 @@
+// kernels.hip — hand-written gfx950 (CDNA4) kernels for the pattern suite.
+//
+// These are the MI355X-native equivalents of the reference's device-code
+// sites: the SYCL busy_wait parallel_for
+// (reference concurency/bench_sycl.cpp:92-97 + bench.hpp:23-31), the A2B copy
+// (bench_sycl.cpp:100), Accumulate / Initialize
+// (allreduce-mpi-sycl.cpp:27-41), and the checksum payload
+// (peer2pear.cpp:8-17). All kernels use 256-thread blocks (4 wave64) and
+// grid-stride loops sized well past the 256 CUs / 8 XCDs.
+
+#include "include/hpk.h"
+
+#include <cstdio>
+#include <stdexcept>
+#include <string>
+
+namespace hpk {
+
+void check_hip(hipError_t e, const char* what) {
+  if (e != hipSuccess) {
+    throw std::runtime_error(std::string("HIP error in ") + what + ": " +
+                             hipGetErrorString(e));
+  }
+}
+
+namespace {
+
+constexpr int kBlock = 256; // 4 wave64 per workgroup
+// Streaming grid cap: 256 CUs want >> 256 workgroups in flight; 16384 blocks
+// of 256 threads = 4.2M threads, plenty for 8 XCDs without oversubscribing
+// the dispatcher.
+constexpr size_t kMaxGrid = 16384;
+
+inline size_t stream_grid(size_t n_items) {
+  size_t blocks = (n_items + kBlock - 1) / kBlock;
+  if (blocks == 0) blocks = 1;
+  if (blocks > kMaxGrid) blocks = kMaxGrid;
+  return blocks;
+}
+
+// --------------------------------------------------------------------------
+// K1: busy-wait FMA chain. Each work-item performs 64*tripcount dependent
+// v_fma_f32 ops on registers and stores one float. The dependency chain makes
+// the kernel's duration linear in tripcount (the property the autotuner's
+// linear regression relies on) and independent of memory traffic.
+// --------------------------------------------------------------------------
+__global__ void k_busy_wait(float* __restrict__ out, long tripcount,
+                            long globalsize) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < globalsize; i += stride) {
+    float x = 1.0f + (float)i * 1e-9f;
+    const float y = 1.000001f;
+    for (long t = 0; t < tripcount; ++t) {
+#pragma unroll
+      for (int k = 0; k < 64; ++k) {
+        x = __builtin_fmaf(y, x, y); // dependent chain: not foldable
+      }
+    }
+    out[i] = x;
+  }
+}
+
+// --------------------------------------------------------------------------
+// K1-MFMA: matrix-core busy loop. Each wave chains `tripcount`
+// v_mfma_f32_16x16x32_bf16 instructions through one accumulator, so the
+// kernel occupies the MFMA pipe (visible as MfmaUtil in rocprof PMC runs)
+// while doing no memory traffic.
+// --------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+__global__ void k_busy_wait_mfma(float* __restrict__ out, long tripcount) {
+  // Arbitrary nonzero fragments; value is irrelevant, occupancy of the MFMA
+  // pipe is the payload.
+  short seed = (short)(threadIdx.x + 1);
+  bf16x8_t a = {seed, seed, seed, seed, seed, seed, seed, seed};
+  bf16x8_t b = {1, 2, 3, 4, 5, 6, 7, 8};
+  f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+  for (long t = 0; t < tripcount; ++t) {
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  // One store per thread keeps the accumulator alive without noise.
+  out[(size_t)blockIdx.x * blockDim.x + threadIdx.x] =
+      acc[0] * 1e-30f; // scaled so huge accumulations don't overflow readers
+}
+
+// --------------------------------------------------------------------------
+// K2: shader copy, 16 B per lane per iteration (1 KiB per wave-instruction).
+// The SDMA sibling is plain hipMemcpyAsync; this kernel measures the
+// shader-blit path and doubles as a D2D bandwidth workload.
+// --------------------------------------------------------------------------
+__global__ void k_copy_b16(const uint4* __restrict__ src, uint4* __restrict__ dst,
+                           size_t n16) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n16; i += stride) dst[i] = src[i];
+}
+
+__global__ void k_copy_b1(const unsigned char* __restrict__ src,
+                          unsigned char* __restrict__ dst, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[i];
+}
+
+// --------------------------------------------------------------------------
+// K4: fills (reference Initialize) and iota payload.
+// --------------------------------------------------------------------------
+__global__ void k_fill_f4(float4* __restrict__ dst, float v, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  float4 val = make_float4(v, v, v, v);
+  for (; i < n4; i += stride) dst[i] = val;
+}
+
+__global__ void k_fill_f1(float* __restrict__ dst, float v, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = v;
+}
+
+__global__ void k_iota_f32(float* __restrict__ dst, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = (float)i;
+}
+
+// --------------------------------------------------------------------------
+// K3: accumulate dst[i] += src[i] (the ring-allreduce reduction kernel).
+// float4 vectorized body + scalar tail.
+// --------------------------------------------------------------------------
+__global__ void k_acc_f4(float4* __restrict__ dst, const float4* __restrict__ src,
+                         size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 d = dst[i];
+    float4 s = src[i];
+    d.x += s.x;
+    d.y += s.y;
+    d.z += s.z;
+    d.w += s.w;
+    dst[i] = d;
+  }
+}
+
+__global__ void k_acc_f1(float* __restrict__ dst, const float* __restrict__ src,
+                         size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] += src[i];
+}
+
+// --------------------------------------------------------------------------
+// Exact checksum: double partial sums per block (wave64 shuffle reduction,
+// then LDS across the block's 4 waves), host-side final sum of <=1024
+// partials.
+// --------------------------------------------------------------------------
+__global__ void k_reduce_partial_f32(const float* __restrict__ src, size_t n,
+                                     double* __restrict__ partial) {
+  double s = 0.0;
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) s += (double)src[i];
+
+  // wave64 reduction
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+
+  __shared__ double wsum[kBlock / 64];
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  if (lane == 0) wsum[wave] = s;
+  __syncthreads();
+  if (wave == 0) {
+    double b = (lane < kBlock / 64) ? wsum[lane] : 0.0;
+    for (int off = 32; off > 0; off >>= 1) b += __shfl_down(b, off, 64);
+    if (lane == 0) partial[blockIdx.x] = b;
+  }
+}
+
+} // namespace
+
+void launch_busy_wait(float* out, long tripcount, long globalsize,
+                      hipStream_t stream) {
+  size_t grid = stream_grid((size_t)globalsize);
+  hipLaunchKernelGGL(k_busy_wait, dim3(grid), dim3(kBlock), 0, stream, out,
+                     tripcount, globalsize);
+  check_hip(hipGetLastError(), "launch_busy_wait");
+}
+
+void launch_busy_wait_mfma(float* out, long tripcount, long n_waves,
+                           hipStream_t stream) {
+  if (n_waves < 1) n_waves = 1;
+  size_t blocks = ((size_t)n_waves * 64 + kBlock - 1) / kBlock;
+  hipLaunchKernelGGL(k_busy_wait_mfma, dim3(blocks), dim3(kBlock), 0, stream,
+                     out, tripcount);
+  check_hip(hipGetLastError(), "launch_busy_wait_mfma");
+}
+
+void launch_copy_kernel(void* dst, const void* src, size_t nbytes,
+                        hipStream_t stream) {
+  uintptr_t d = (uintptr_t)dst, s = (uintptr_t)src;
+  if ((d % 16 == 0) && (s % 16 == 0)) {
+    size_t n16 = nbytes / 16;
+    size_t tail = nbytes - n16 * 16;
+    if (n16) {
+      hipLaunchKernelGGL(k_copy_b16, dim3(stream_grid(n16)), dim3(kBlock), 0,
+                         stream, (const uint4*)src, (uint4*)dst, n16);
+    }
+    if (tail) {
+      hipLaunchKernelGGL(k_copy_b1, dim3(1), dim3(kBlock), 0, stream,
+                         (const unsigned char*)src + n16 * 16,
+                         (unsigned char*)dst + n16 * 16, tail);
+    }
+  } else {
+    hipLaunchKernelGGL(k_copy_b1, dim3(stream_grid(nbytes)), dim3(kBlock), 0,
+                       stream, (const unsigned char*)src, (unsigned char*)dst,
+                       nbytes);
+  }
+  check_hip(hipGetLastError(), "launch_copy_kernel");
+}
+
+void launch_fill_f32(float* dst, float value, size_t n, hipStream_t stream) {
+  if (((uintptr_t)dst % 16 == 0) && n >= 4) {
+    size_t n4 = n / 4;
+    size_t tail = n - n4 * 4;
+    hipLaunchKernelGGL(k_fill_f4, dim3(stream_grid(n4)), dim3(kBlock), 0,
+                       stream, (float4*)dst, value, n4);
+    if (tail) {
+      hipLaunchKernelGGL(k_fill_f1, dim3(1), dim3(64), 0, stream, dst + n4 * 4,
+                         value, tail);
+    }
+  } else {
+    hipLaunchKernelGGL(k_fill_f1, dim3(stream_grid(n)), dim3(kBlock), 0, stream,
+                       dst, value, n);
+  }
+  check_hip(hipGetLastError(), "launch_fill_f32");
+}
+
+void launch_iota_f32(float* dst, size_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_iota_f32, dim3(stream_grid(n)), dim3(kBlock), 0, stream,
+                     dst, n);
+  check_hip(hipGetLastError(), "launch_iota_f32");
+}
+
+void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream) {
+  if (((uintptr_t)dst % 16 == 0) && ((uintptr_t)src % 16 == 0) && n >= 4) {
+    size_t n4 = n / 4;
+    size_t tail = n - n4 * 4;
+    hipLaunchKernelGGL(k_acc_f4, dim3(stream_grid(n4)), dim3(kBlock), 0, stream,
+                       (float4*)dst, (const float4*)src, n4);
+    if (tail) {
+      hipLaunchKernelGGL(k_acc_f1, dim3(1), dim3(64), 0, stream, dst + n4 * 4,
+                         src + n4 * 4, tail);
+    }
+  } else {
+    hipLaunchKernelGGL(k_acc_f1, dim3(stream_grid(n)), dim3(kBlock), 0, stream,
+                       dst, src, n);
+  }
+  check_hip(hipGetLastError(), "launch_acc_f32");
+}
+
+double reduce_sum_f32(const float* src, size_t n, hipStream_t stream) {
+  constexpr size_t kRedBlocks = 1024;
+  size_t blocks = (n + (size_t)kBlock * 8 - 1) / ((size_t)kBlock * 8);
+  if (blocks == 0) blocks = 1;
+  if (blocks > kRedBlocks) blocks = kRedBlocks;
+
+  double* d_partial = nullptr;
+  check_hip(hipMallocAsync((void**)&d_partial, blocks * sizeof(double), stream),
+            "reduce_sum_f32 hipMallocAsync");
+  hipLaunchKernelGGL(k_reduce_partial_f32, dim3(blocks), dim3(kBlock), 0,
+                     stream, src, n, d_partial);
+  check_hip(hipGetLastError(), "reduce_sum_f32 kernel");
+
+  std::vector<double> h(blocks);
+  check_hip(hipMemcpyAsync(h.data(), d_partial, blocks * sizeof(double),
+                           hipMemcpyDeviceToHost, stream),
+            "reduce_sum_f32 D2H");
+  check_hip(hipFreeAsync(d_partial, stream), "reduce_sum_f32 hipFreeAsync");
+  check_hip(hipStreamSynchronize(stream), "reduce_sum_f32 sync");
+
+  double s = 0.0;
+  for (double v : h) s += v;
+  return s;
+}
+
+} // namespace hpk
