@@ -1,0 +1,138 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark of the MI355X pattern suite.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+N>1 is launched as `python -m torch.distributed.run --nnodes=1
+--nproc-per-node N ... bench.py` (one rank per GPU over RCCL); rank/world
+come from the environment. W untimed warmup steps, then exactly K timed
+steps bracketed by barrier + torch.cuda.synchronize on both sides, MAX over
+ranks, one JSON line from rank 0.
+
+The step (models/flagship.py) bundles the suite's headline patterns —
+compute/copy stream overlap, pairwise xGMI P2P exchange, RCCL all-reduce —
+and the value is whole-job aggregate payload bandwidth in GB/s
+(higher-is-better; BASELINE.json metric components reported alongside).
+The reference publishes no absolute numbers, so vs_baseline
+is null.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--smoke", action="store_true",
+                    help="tiny config (seconds, used by __graft_entry__)")
+    args = ap.parse_args()
+
+    import torch
+
+    if not torch.cuda.is_available():
+        print("bench.py requires a GPU (MI355X)", file=sys.stderr)
+        return 1
+
+    import torch.distributed as dist
+
+    from hpc_patterns_amd.models import SMOKE_CONFIG, FlagshipPatternStep
+    from hpc_patterns_amd.models.flagship import DEFAULT_CONFIG
+    from hpc_patterns_amd.parallel import init_distributed
+    from hpc_patterns_amd.parallel.p2p import pairwise_bandwidth, pingpong
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    if world > 1:
+        rank, local_rank, world = init_distributed()
+    else:
+        rank, local_rank = 0, 0
+        torch.cuda.set_device(0)
+
+    device = torch.device("cuda", torch.cuda.current_device())
+    cfg = dict(SMOKE_CONFIG if args.smoke else DEFAULT_CONFIG)
+    step = FlagshipPatternStep(device=device, rank=rank, world_size=world,
+                               config=cfg)
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step.step()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step.step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    bytes_per_step = step.bytes_per_step_per_rank() * world
+    agg_gbps = bytes_per_step * args.steps / elapsed / 1e9
+    ms_per_step = elapsed / args.steps * 1e3
+
+    # ---- component diagnostics (outside the timed region) ----
+    components = {}
+    overlap = step.measure_overlap(reps=3)
+    components["stream_overlap_pct"] = round(
+        100.0 * overlap["overlap_efficiency"], 1)
+    components["stream_overlap_speedup"] = round(overlap["speedup"], 3)
+    components["theoretical_speedup"] = round(
+        overlap["theoretical_speedup"], 3)
+    if world > 1:
+        bw = pairwise_bandwidth(cfg["p2p_floats"] * 4, iters=5,
+                                bidirectional=False, device=device)
+        components["p2p_uni_GBps"] = round(bw["gbps"], 2)
+        components["p2p_checksum_ok"] = bool(bw["checksum_ok"])
+        pp = pingpong(nbytes=8, iters=50, device=device)
+        components["pingpong_us"] = round(pp["oneway_us"], 2)
+
+    if rank == 0:
+        result = {
+            "metric": "pattern_aggregate_GBps",
+            "value": round(agg_gbps, 2),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (random-init/iota payloads, reference sizes)",
+            "config": {
+                "model": "flagship-pattern-step",
+                "d2d_bytes": cfg["d2d_floats"] * 4,
+                "h2d_bytes": cfg["h2d_bytes"],
+                "d2h_bytes": cfg["d2h_bytes"],
+                "p2p_bytes": cfg["p2p_floats"] * 4,
+                "allreduce_bytes": cfg["allreduce_floats"] * 4,
+                "tripcount_C": cfg["tripcount"],
+                "parallelism": f"dp{world}",
+            },
+            "components": components,
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,218 @@
+"""FlagshipPatternStep — the benchmark step of the suite.
+
+One "step" per GPU, all submitted concurrently on separate hipStreams
+(reference concurrency pattern), then joined:
+
+  stream 0: K1 busy-wait compute kernel (calibrated to the copy time)
+  stream 1: K2 shader D2D copy (1 GiB payload by default)
+  stream 2: H2D hipMemcpyAsync from pinned host memory
+  stream 3: D2H hipMemcpyAsync to pinned host memory
+  world>1 : all-reduce of 2^25 floats (RCCL) + pairwise 188.7 MB P2P
+            exchange (ncclSend/Recv over xGMI) — the reference's C4/C5/C1
+            communication patterns
+
+The benchmark value is whole-job aggregate bandwidth: payload bytes moved by
+all ranks divided by step time (max over ranks). Compute contributes no
+bytes — it is there to prove copies and collectives overlap compute, which
+is the reference's headline criterion (speedup within 30% of theoretical,
+main.cpp:12).
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+
+import torch
+import torch.distributed as dist
+
+# Reference workload sizes: 1 GB copy commands (concurency/main.cpp:100-105),
+# 47,185,920-float P2P pairs (peer2pear.cpp:115), 2^25-float allreduce
+# (allreduce-mpi-sycl.cpp:99).
+DEFAULT_CONFIG = dict(
+    d2d_floats=1 << 28,        # 1 GiB shader-copy payload
+    h2d_bytes=256 << 20,       # 256 MiB pinned H2D
+    d2h_bytes=256 << 20,       # 256 MiB pinned D2H
+    p2p_floats=47_185_920,     # 188.7 MB pairwise exchange
+    allreduce_floats=1 << 25,  # 128 MiB all-reduce
+    tripcount=-1,              # autotuned to match the slowest copy
+    compute_globalsize=1 << 20,
+)
+
+SMOKE_CONFIG = dict(
+    d2d_floats=1 << 20,
+    h2d_bytes=1 << 20,
+    d2h_bytes=1 << 20,
+    p2p_floats=1 << 18,
+    allreduce_floats=1 << 18,
+    tripcount=64,
+    compute_globalsize=1 << 14,
+)
+
+
+@dataclass
+class FlagshipPatternStep:
+    device: torch.device
+    rank: int = 0
+    world_size: int = 1
+    config: dict = field(default_factory=lambda: dict(DEFAULT_CONFIG))
+    use_distributed: bool = True
+
+    def __post_init__(self):
+        from .. import ops
+
+        self.ops = ops
+        cfg = self.config
+        dev = self.device
+        torch.cuda.set_device(dev)
+        self.streams = [torch.cuda.Stream(device=dev) for _ in range(4)]
+
+        # buffers
+        self.d2d_src = torch.empty(cfg["d2d_floats"], dtype=torch.float32, device=dev)
+        self.d2d_dst = torch.empty_like(self.d2d_src)
+        ops.fill(self.d2d_src, 1.0)
+        self.h2d_host = torch.empty(cfg["h2d_bytes"] // 4, dtype=torch.float32,
+                                    pin_memory=True)
+        self.h2d_dev = torch.empty_like(self.h2d_host, device=dev)
+        self.d2h_dev = torch.empty(cfg["d2h_bytes"] // 4, dtype=torch.float32,
+                                   device=dev)
+        ops.fill(self.d2h_dev, 2.0)
+        self.d2h_host = torch.empty_like(self.d2h_dev, device="cpu",
+                                         pin_memory=True)
+        self.compute_out = torch.empty(cfg["compute_globalsize"],
+                                       dtype=torch.float32, device=dev)
+
+        self.distributed = (self.use_distributed and self.world_size > 1
+                            and dist.is_initialized())
+        if self.distributed:
+            self.ar_buf = torch.ones(cfg["allreduce_floats"], dtype=torch.float32,
+                                     device=dev)
+            self.p2p_send = torch.empty(cfg["p2p_floats"], dtype=torch.float32,
+                                        device=dev)
+            ops.iota(self.p2p_send)
+            self.p2p_recv = torch.empty_like(self.p2p_send)
+            self.peer = self._pair_peer()
+
+        torch.cuda.synchronize()
+        if cfg["tripcount"] == -1:
+            cfg["tripcount"] = self._calibrate_tripcount()
+
+    def _pair_peer(self):
+        peer = self.rank + 1 if self.rank % 2 == 0 else self.rank - 1
+        return peer if peer < self.world_size else None
+
+    # ---- calibration: make the compute command last about as long as the
+    # slowest copy (linear model, reference autotuner main.cpp:226-258) ----
+    def _calibrate_tripcount(self, probe: int = 200) -> int:
+        dev_sync = torch.cuda.synchronize
+        t0 = time.perf_counter()
+        self.ops.copy_kernel(self.d2d_dst, self.d2d_src)
+        dev_sync()
+        t_d2d = time.perf_counter() - t0
+        t0 = time.perf_counter()
+        self.h2d_dev.copy_(self.h2d_host, non_blocking=True)
+        dev_sync()
+        t_h2d = time.perf_counter() - t0
+        target = max(t_d2d, t_h2d)
+
+        t0 = time.perf_counter()
+        self.ops.busy_wait(self.compute_out, probe,
+                           self.config["compute_globalsize"])
+        dev_sync()
+        t_probe = max(time.perf_counter() - t0, 1e-7)
+        return max(int(probe * target / t_probe), 1)
+
+    # ---- the step ----
+    def step(self) -> None:
+        cfg = self.config
+        s0, s1, s2, s3 = self.streams
+        with torch.cuda.stream(s0):
+            self.ops.busy_wait(self.compute_out, cfg["tripcount"],
+                               cfg["compute_globalsize"], stream=s0)
+        with torch.cuda.stream(s1):
+            self.ops.copy_kernel(self.d2d_dst, self.d2d_src, stream=s1)
+        with torch.cuda.stream(s2):
+            self.h2d_dev.copy_(self.h2d_host, non_blocking=True)
+        with torch.cuda.stream(s3):
+            self.d2h_host.copy_(self.d2h_dev, non_blocking=True)
+
+        if self.distributed:
+            # collectives ride torch's comm stream, overlapping the local work
+            ar_work = dist.all_reduce(self.ar_buf, async_op=True)
+            if self.peer is not None:
+                reqs = dist.batch_isend_irecv([
+                    dist.P2POp(dist.isend, self.p2p_send, self.peer),
+                    dist.P2POp(dist.irecv, self.p2p_recv, self.peer),
+                ])
+                for r in reqs:
+                    r.wait()
+            ar_work.wait()
+
+        for s in self.streams:
+            s.synchronize()
+
+    # ---- accounting ----
+    def bytes_per_step_per_rank(self) -> int:
+        cfg = self.config
+        b = cfg["d2d_floats"] * 4 + cfg["h2d_bytes"] + cfg["d2h_bytes"]
+        if self.distributed:
+            if self.peer is not None:
+                b += 2 * cfg["p2p_floats"] * 4  # send + recv payload
+            # ring-equivalent bus bytes per rank for an allreduce
+            n = self.world_size
+            b += int(2 * (n - 1) / n * cfg["allreduce_floats"] * 4)
+        return b
+
+    # ---- overlap diagnostic (the reference speedup criterion, local only) --
+    def measure_overlap(self, reps: int = 5) -> dict:
+        """Serial vs concurrent submission of the 4 local commands."""
+        cfg = self.config
+
+        def serial_once():
+            times = []
+            for fn in (
+                lambda: self.ops.busy_wait(self.compute_out, cfg["tripcount"],
+                                           cfg["compute_globalsize"]),
+                lambda: self.ops.copy_kernel(self.d2d_dst, self.d2d_src),
+                lambda: self.h2d_dev.copy_(self.h2d_host, non_blocking=True),
+                lambda: self.d2h_host.copy_(self.d2h_dev, non_blocking=True),
+            ):
+                t0 = time.perf_counter()
+                fn()
+                torch.cuda.synchronize()
+                times.append(time.perf_counter() - t0)
+            return times
+
+        def concurrent_once():
+            s0, s1, s2, s3 = self.streams
+            t0 = time.perf_counter()
+            with torch.cuda.stream(s0):
+                self.ops.busy_wait(self.compute_out, cfg["tripcount"],
+                                   cfg["compute_globalsize"], stream=s0)
+            with torch.cuda.stream(s1):
+                self.ops.copy_kernel(self.d2d_dst, self.d2d_src, stream=s1)
+            with torch.cuda.stream(s2):
+                self.h2d_dev.copy_(self.h2d_host, non_blocking=True)
+            with torch.cuda.stream(s3):
+                self.d2h_host.copy_(self.d2h_dev, non_blocking=True)
+            for s in self.streams:
+                s.synchronize()
+            return time.perf_counter() - t0
+
+        serial_best, per_cmd_best = float("inf"), None
+        conc_best = float("inf")
+        for _ in range(reps):
+            times = serial_once()
+            if sum(times) < serial_best:
+                serial_best, per_cmd_best = sum(times), times
+            conc_best = min(conc_best, concurrent_once())
+        theoretical = serial_best / max(per_cmd_best)
+        speedup = serial_best / conc_best
+        return {
+            "serial_s": serial_best,
+            "concurrent_s": conc_best,
+            "per_command_s": per_cmd_best,
+            "theoretical_speedup": theoretical,
+            "speedup": speedup,
+            "overlap_efficiency": speedup / theoretical,
+        }

@@ -1,0 +1,104 @@
+"""ops — hand-written HIP/CDNA4 kernels exposed on torch tensors.
+
+Python face of the native kernels in native/kernels.hip (the MI355X
+equivalents of the reference's device-code sites). All
+functions launch on the CURRENT torch CUDA stream unless given one, and all
+REQUIRE the native extension when a GPU is present — no silent eager
+fallback.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .._native import native
+
+
+def _stream_handle(stream=None) -> int:
+    if stream is not None:
+        return stream.cuda_stream if hasattr(stream, "cuda_stream") else int(stream)
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _check_f32(t: torch.Tensor, name: str):
+    if t.dtype != torch.float32 or not t.is_cuda or not t.is_contiguous():
+        raise TypeError(f"{name} must be a contiguous float32 CUDA tensor")
+
+
+def busy_wait(out: torch.Tensor, tripcount: int, globalsize: int | None = None,
+              stream=None) -> None:
+    """K1: every work-item runs 64*tripcount dependent FMAs, stores 1 float.
+
+    Reference payload: concurency/bench.hpp:23-31 (MAD_64) submitted at
+    bench_sycl.cpp:92-97.
+    """
+    _check_f32(out, "out")
+    gs = out.numel() if globalsize is None else globalsize
+    if gs > out.numel():
+        raise ValueError("globalsize exceeds out buffer")
+    native().busy_wait(out.data_ptr(), int(tripcount), int(gs),
+                       _stream_handle(stream))
+
+
+def busy_wait_mfma(out: torch.Tensor, tripcount: int, n_waves: int = 1,
+                   stream=None) -> None:
+    """K1-MFMA: chained v_mfma_f32_16x16x32_bf16 busy loop (matrix cores)."""
+    _check_f32(out, "out")
+    if out.numel() < ((n_waves * 64 + 255) // 256) * 256:
+        raise ValueError("out too small: need >= ceil(n_waves*64/256)*256 floats")
+    native().busy_wait_mfma(out.data_ptr(), int(tripcount), int(n_waves),
+                            _stream_handle(stream))
+
+
+def copy_kernel(dst: torch.Tensor, src: torch.Tensor, stream=None) -> None:
+    """K2: vectorized shader copy (the blit-engine sibling of hipMemcpyAsync)."""
+    if dst.numel() * dst.element_size() != src.numel() * src.element_size():
+        raise ValueError("size mismatch")
+    if not (dst.is_contiguous() and src.is_contiguous()):
+        raise TypeError("contiguous tensors required")
+    native().copy_kernel(dst.data_ptr(), src.data_ptr(),
+                         dst.numel() * dst.element_size(),
+                         _stream_handle(stream))
+
+
+def fill(dst: torch.Tensor, value: float, stream=None) -> None:
+    """K4: dst[:] = value (reference Initialize, allreduce-mpi-sycl.cpp:34-41)."""
+    _check_f32(dst, "dst")
+    native().fill_f32(dst.data_ptr(), float(value), dst.numel(),
+                      _stream_handle(stream))
+
+
+def iota(dst: torch.Tensor, stream=None) -> None:
+    """dst[i] = float(i) — device-side checksum payload generator."""
+    _check_f32(dst, "dst")
+    native().iota_f32(dst.data_ptr(), dst.numel(), _stream_handle(stream))
+
+
+def accumulate(dst: torch.Tensor, src: torch.Tensor, stream=None) -> None:
+    """K3: dst += src (reference Accumulate, allreduce-mpi-sycl.cpp:27-31)."""
+    _check_f32(dst, "dst")
+    _check_f32(src, "src")
+    if dst.numel() != src.numel():
+        raise ValueError("size mismatch")
+    native().acc_f32(dst.data_ptr(), src.data_ptr(), dst.numel(),
+                     _stream_handle(stream))
+
+
+def reduce_sum(src: torch.Tensor, stream=None) -> float:
+    """Exact float64 sum of a float32 tensor (device reduction; synchronizes).
+
+    Order-independent when all addends are integer-valued (every partial sum
+    stays below 2^53), which is what the checksum payloads guarantee —
+    replaces the reference's host sort+sum (peer2pear.cpp:56-63).
+    """
+    _check_f32(src, "src")
+    return native().reduce_sum_f32(src.data_ptr(), src.numel(),
+                                   _stream_handle(stream))
+
+
+def iota_checksum(n: int) -> float:
+    """Expected exact double sum of [float(i) for i in range(n)]."""
+    import numpy as np
+
+    # float32 rounding of the iota values, summed exactly in float64
+    return float(np.arange(n, dtype=np.float32).astype(np.float64).sum())

@@ -1,0 +1,137 @@
+"""Hand-written ring all-reduce patterns over torch.distributed pt2pt.
+
+MI355X-native re-design of the reference ring-allreduce miniapps
+(reference allreduce-mpi-sycl.cpp:44-59,173-182 SendRecvRing + Accumulate):
+blocking MPI_Send/Recv becomes dist.batch_isend_irecv (ncclSend/ncclRecv over
+xGMI on GPU, gloo on CPU — concurrent send+recv, so the reference's
+odd/even-send-first deadlock dance is unnecessary), and the accumulate is the
+hand-written HIP K3 kernel on GPU.
+
+Three variants:
+- ring_allreduce:            the reference pattern — full buffer circulates
+                             (size-1) times; per-link traffic (size-1)*bytes.
+- ring_allreduce_pipelined:  chunked, transfer/accumulate overlapped — the
+                             tuned variant the reference leaves as future
+                             work.
+- ring_allreduce_rsag:       bandwidth-optimal reduce-scatter + all-gather
+                             ring (2(size-1)/size * bytes per link) — the
+                             "beat the reference" variant.
+
+All are verified against dist.all_reduce (RCCL) by the test-suite's analytic
+oracle: fill(rank) -> every element == size*(size-1)/2.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+
+def _acc(dst: torch.Tensor, src: torch.Tensor) -> None:
+    if dst.is_cuda:
+        from .. import ops
+
+        ops.accumulate(dst, src)
+    else:
+        dst.add_(src)
+
+
+def _ring_neighbours(group=None) -> tuple[int, int, int, int]:
+    rank = dist.get_rank(group)
+    size = dist.get_world_size(group)
+    right = (rank + 1) % size
+    left = (rank - 1 + size) % size
+    if group is not None and group is not dist.group.WORLD:
+        right = dist.get_global_rank(group, right)
+        left = dist.get_global_rank(group, left)
+    return rank, size, right, left
+
+
+def ring_allreduce(tensor: torch.Tensor, group=None) -> torch.Tensor:
+    """In-place sum-allreduce via the naive full-buffer ring."""
+    rank, size, right, left = _ring_neighbours(group)
+    if size == 1:
+        return tensor
+    send = tensor.clone()
+    recv = torch.empty_like(tensor)
+    for _ in range(size - 1):
+        ops_ = [
+            dist.P2POp(dist.isend, send, right, group),
+            dist.P2POp(dist.irecv, recv, left, group),
+        ]
+        for req in dist.batch_isend_irecv(ops_):
+            req.wait()
+        _acc(tensor, recv)
+        send, recv = recv, send
+    return tensor
+
+
+def ring_allreduce_pipelined(tensor: torch.Tensor, group=None,
+                             n_chunks: int = 8) -> torch.Tensor:
+    """Chunked ring: chunk c's accumulate overlaps chunk c+1's transfer."""
+    rank, size, right, left = _ring_neighbours(group)
+    if size == 1:
+        return tensor
+    flat = tensor.view(-1)
+    chunks = list(torch.chunk(flat, min(n_chunks, max(flat.numel(), 1))))
+    send_buf = flat.clone()
+    recv_buf = torch.empty_like(flat)
+    send_chunks = list(torch.chunk(send_buf, len(chunks)))
+    recv_chunks = list(torch.chunk(recv_buf, len(chunks)))
+    for _ in range(size - 1):
+        reqs = []
+        for sc, rc in zip(send_chunks, recv_chunks):
+            reqs.append(dist.batch_isend_irecv([
+                dist.P2POp(dist.isend, sc, right, group),
+                dist.P2POp(dist.irecv, rc, left, group),
+            ]))
+        for c, chunk_reqs in enumerate(reqs):
+            for req in chunk_reqs:
+                req.wait()
+            _acc(chunks[c], recv_chunks[c])
+        send_buf, recv_buf = recv_buf, send_buf
+        send_chunks, recv_chunks = recv_chunks, send_chunks
+    return tensor
+
+
+def ring_allreduce_rsag(tensor: torch.Tensor, group=None) -> torch.Tensor:
+    """Bandwidth-optimal ring: reduce-scatter pass then all-gather pass.
+
+    Per-rank traffic 2*(size-1)/size*bytes vs the naive ring's
+    (size-1)*bytes. Requires numel divisible by world size (pad upstream if
+    needed)."""
+    rank, size, right, left = _ring_neighbours(group)
+    if size == 1:
+        return tensor
+    flat = tensor.view(-1)
+    if flat.numel() % size != 0:
+        raise ValueError("numel must be divisible by world size "
+                         f"({flat.numel()} % {size} != 0)")
+    parts = list(torch.chunk(flat, size))
+    scratch = torch.empty_like(parts[0])
+
+    # reduce-scatter: after step s, rank owns the full sum of part
+    # (rank+1) mod size ... progressing to part rank.
+    for step in range(size - 1):
+        send_idx = (rank - step) % size
+        recv_idx = (rank - step - 1) % size
+        reqs = dist.batch_isend_irecv([
+            dist.P2POp(dist.isend, parts[send_idx].contiguous(), right, group),
+            dist.P2POp(dist.irecv, scratch, left, group),
+        ])
+        for req in reqs:
+            req.wait()
+        _acc(parts[recv_idx], scratch)
+
+    # all-gather: circulate the owned (fully reduced) parts.
+    for step in range(size - 1):
+        send_idx = (rank - step + 1) % size
+        recv_idx = (rank - step) % size
+        reqs = dist.batch_isend_irecv([
+            dist.P2POp(dist.isend, parts[send_idx].contiguous(), right, group),
+            dist.P2POp(dist.irecv, scratch, left, group),
+        ])
+        for req in reqs:
+            req.wait()
+        parts[recv_idx].copy_(scratch)
+    return tensor

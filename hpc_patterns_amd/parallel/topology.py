@@ -1,0 +1,84 @@
+"""xGMI topology discovery (python face).
+
+Wraps the native link matrix (native/topo.hip: hipDeviceCanAccessPeer +
+hipExtGetLinkTypeAndHopCount + rocm_smi weights/bandwidths) — the MI355X
+replacement of the reference Level-Zero-Sysman fabric enumerator
+(reference p2p/topology.cpp). Pure helpers (plane merge, topology order)
+are CPU-testable on synthetic matrices.
+"""
+
+from __future__ import annotations
+
+LINK_TYPE_XGMI = 2  # hipExtGetLinkTypeAndHopCount: 2 == HSA_AMD_LINK_INFO_TYPE_XGMI
+
+
+def link_matrix() -> list[list[dict]]:
+    """NxN link matrix from the native core (requires GPU)."""
+    from .._native import native
+
+    return native().link_matrix()
+
+
+def planes_from_matrix(matrix: list[list[dict]]) -> list[list[int]]:
+    """Connected components under direct-P2P reachability (pure python;
+    the reference's connectivity-plane merge, topology.cpp:76-89)."""
+    n = len(matrix)
+    parent = list(range(n))
+
+    def find(x: int) -> int:
+        while parent[x] != x:
+            parent[x] = parent[parent[x]]
+            x = parent[x]
+        return x
+
+    for i in range(n):
+        for j in range(n):
+            if i != j and matrix[i][j].get("p2p"):
+                a, b = find(i), find(j)
+                if a != b:
+                    parent[a] = b
+    planes: dict[int, list[int]] = {}
+    for i in range(n):
+        planes.setdefault(find(i), []).append(i)
+    return list(planes.values())
+
+
+def topology_order_from_matrix(matrix: list[list[dict]]) -> list[int]:
+    """Greedy walk: start at GPU 0, repeatedly hop to the unvisited neighbour
+    with the highest link weight (fewest hops as tiebreak), so consecutive
+    positions are directly connected. Falls back to numeric order for
+    disconnected leftovers."""
+    n = len(matrix)
+    if n == 0:
+        return []
+    order = [0]
+    visited = {0}
+    while len(order) < n:
+        cur = order[-1]
+        best, best_key = None, None
+        for j in range(n):
+            if j in visited:
+                continue
+            li = matrix[cur][j]
+            if not li.get("p2p"):
+                continue
+            # higher weight better; fewer hops better; stable by index
+            key = (-(li.get("weight") or 0), li.get("hops") or 0, j)
+            if best_key is None or key < best_key:
+                best, best_key = j, key
+        if best is None:  # disconnected: take smallest unvisited
+            best = min(set(range(n)) - visited)
+        order.append(best)
+        visited.add(best)
+    return order
+
+
+def topology_order(n_gpus: int) -> list[int]:
+    """Topology order from live hardware; numeric order if discovery fails."""
+    try:
+        m = link_matrix()
+        if len(m) >= n_gpus:
+            return topology_order_from_matrix(m)[:n_gpus]
+    except Exception:
+        pass
+    return list(range(n_gpus))

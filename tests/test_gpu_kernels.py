@@ -1,0 +1,131 @@
+"""GPU numerics tests: every hand-written HIP kernel vs a plain PyTorch fp32
+reference (SURVEY.md §4 oracle style — exact where the math is exact)."""
+
+import pytest
+import torch
+
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(600)]
+
+
+@pytest.fixture(scope="module")
+def dev():
+    torch.cuda.set_device(0)
+    return torch.device("cuda", 0)
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from hpc_patterns_amd import ops as _ops
+    from hpc_patterns_amd._native import native
+
+    native()  # fail loudly, not skip: on a GPU box the extension must exist
+    return _ops
+
+
+@pytest.mark.parametrize("n", [1, 3, 4, 255, 1 << 20, (1 << 20) + 7])
+def test_fill(ops, dev, n):
+    t = torch.empty(n, dtype=torch.float32, device=dev)
+    ops.fill(t, 3.5)
+    torch.cuda.synchronize()
+    assert torch.equal(t, torch.full((n,), 3.5, device=dev))
+
+
+@pytest.mark.parametrize("n", [1, 17, 1 << 20])
+def test_iota(ops, dev, n):
+    t = torch.empty(n, dtype=torch.float32, device=dev)
+    ops.iota(t)
+    torch.cuda.synchronize()
+    assert torch.equal(t, torch.arange(n, dtype=torch.float32, device=dev))
+
+
+@pytest.mark.parametrize("n", [4, 1023, 1 << 22, (1 << 22) + 3])
+def test_accumulate(ops, dev, n):
+    a = torch.rand(n, device=dev)
+    b = torch.rand(n, device=dev)
+    ref = a + b
+    ops.accumulate(a, b)
+    torch.cuda.synchronize()
+    assert torch.equal(a, ref)  # fp32 add is exact vs torch's fp32 add
+
+
+@pytest.mark.parametrize("nbytes", [16, 4096, (1 << 22) + 36])
+def test_copy_kernel(ops, dev, nbytes):
+    n = nbytes // 4
+    src = torch.rand(n, device=dev)
+    dst = torch.empty_like(src)
+    ops.copy_kernel(dst, src)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, src)
+
+
+def test_copy_kernel_unaligned_tail(ops, dev):
+    base = torch.rand(1029, device=dev)  # odd length -> 16B body + tail
+    dst = torch.empty_like(base)
+    ops.copy_kernel(dst, base)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, base)
+
+
+@pytest.mark.parametrize("n", [1, 1000, 1 << 24])
+def test_reduce_sum_exact_iota(ops, dev, n):
+    t = torch.empty(n, dtype=torch.float32, device=dev)
+    ops.iota(t)
+    got = ops.reduce_sum(t)
+    expected = ops.iota_checksum(n)
+    assert got == expected  # exact: integer-valued addends in double
+
+
+def test_reduce_sum_vs_torch(ops, dev):
+    t = torch.rand(1 << 20, device=dev)
+    got = ops.reduce_sum(t)
+    ref = float(t.to(torch.float64).sum())
+    assert abs(got - ref) < 1e-6 * max(abs(ref), 1.0)
+
+
+def test_busy_wait_runs_and_scales(ops, dev):
+    import time
+
+    out = torch.empty(256, dtype=torch.float32, device=dev)
+    # correctness: writes a finite value everywhere in [0, globalsize)
+    ops.busy_wait(out, tripcount=10, globalsize=256)
+    torch.cuda.synchronize()
+    assert torch.isfinite(out).all()
+
+    def timed(trip):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        ops.busy_wait(out, tripcount=trip, globalsize=256)
+        torch.cuda.synchronize()
+        return time.perf_counter() - t0
+
+    timed(1000)  # warm
+    t1 = min(timed(2000) for _ in range(3))
+    t4 = min(timed(8000) for _ in range(3))
+    # linearity within 2x slack (the autotuner's model)
+    assert 2.0 < t4 / t1 < 8.0, (t1, t4)
+
+
+def test_busy_wait_mfma_runs(ops, dev):
+    out = torch.empty(256, dtype=torch.float32, device=dev)
+    ops.busy_wait_mfma(out, tripcount=1000, n_waves=4)
+    torch.cuda.synchronize()
+    assert torch.isfinite(out).all()
+
+
+def test_ops_reject_wrong_dtype(ops, dev):
+    t = torch.empty(8, dtype=torch.float64, device=dev)
+    with pytest.raises(TypeError):
+        ops.fill(t, 0.0)
+
+
+def test_interop_torch_allocator_shared_with_hip_kernels(ops, dev):
+    """The suite's runtime-interop capability (reference
+    sycl_omp_ze_interopt): torch's caching allocator, torch streams and the
+    raw HIP kernels drive the same memory with no copies."""
+    s = torch.cuda.Stream()
+    t = torch.zeros(1 << 16, dtype=torch.float32, device=dev)
+    with torch.cuda.stream(s):
+        ops.fill(t, 7.0, stream=s)        # native kernel on torch stream
+        u = t * 2.0                        # torch op, same stream, same memory
+    s.synchronize()
+    assert torch.equal(u, torch.full_like(t, 14.0))
