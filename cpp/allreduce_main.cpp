@@ -34,6 +34,7 @@
 //   --algo ring|pipeline|rccl
 
 #include "../hpc_patterns_amd/native/include/hpk.h"
+#include "launch_util.h"
 
 #include <rccl/rccl.h>
 #include <sys/mman.h>
@@ -161,13 +162,14 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
   return dt;
 }
 
-int worker(int rank, int size, int ndev, const ncclUniqueId& id,
-           const Config& cfg, double* shared_out) {
+int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
+           const Config& cfg) {
   int dev = rank % ndev;
   hpk::check_hip(hipSetDevice(dev), "hipSetDevice");
+  ncclUniqueId id;
+  hpk_launch::bootstrap_id(sh, rank, &id);
   ncclComm_t comm;
-  check_nccl(ncclCommInitRank(&comm, size, const_cast<ncclUniqueId&>(id), rank),
-             "ncclCommInitRank");
+  check_nccl(ncclCommInitRank(&comm, size, id, rank), "ncclCommInitRank");
 
   size_t n = 1ull << cfg.p;
   size_t bytes = n * sizeof(float);
@@ -232,7 +234,7 @@ int worker(int rank, int size, int ndev, const ncclUniqueId& id,
     std::printf("# algo=%s ranks=%d elems=2^%d alloc=%c time=%.6fs "
                 "busbw=%.2f GB/s\n",
                 cfg.algo.c_str(), size, cfg.p, cfg.alloc, max_time, busbw);
-    if (shared_out) *shared_out = max_time;
+    if (sh) sh->result = max_time;
   }
 
   ncclCommDestroy(comm);
@@ -250,7 +252,10 @@ int main(int argc, char* argv[]) {
       if (++i >= argc) { std::fprintf(stderr, "missing value\n"); std::exit(1); }
       return argv[i];
     };
-    if (s == "-p") cfg.p = std::atoi(next());
+    if (s == "--probe-ndev") {  // launch_util.h re-exec probe
+      std::printf("%d\n", hpk::device_count());
+      return 0;
+    } else if (s == "-p") cfg.p = std::atoi(next());
     else if (s == "-D") cfg.alloc = 'D';
     else if (s == "-H") cfg.alloc = 'H';
     else if (s == "-S") cfg.alloc = 'S';
@@ -271,7 +276,9 @@ int main(int argc, char* argv[]) {
     return 1;
   }
 
-  int ndev = hpk::device_count();
+  // Parent stays HIP-free (launch_util.h invariant): probe device count by
+  // re-exec, bootstrap the ncclUniqueId inside rank 0 after fork.
+  int ndev = hpk_launch::probe_device_count(argv[0]);
   if (ndev == 0) {
     std::fprintf(stderr, "no HIP devices\n");
     return 1;
@@ -286,27 +293,7 @@ int main(int argc, char* argv[]) {
     size = ndev;
   }
 
-  ncclUniqueId id;
-  check_nccl(ncclGetUniqueId(&id), "ncclGetUniqueId");
-
-  // shared page for rank0's timing result (read by the parent after waitpid)
-  double* shared_out =
-      (double*)mmap(nullptr, sizeof(double), PROT_READ | PROT_WRITE,
-                    MAP_SHARED | MAP_ANONYMOUS, -1, 0);
-
-  if (size == 1) return worker(0, 1, ndev, id, cfg, shared_out);
-
-  std::vector<pid_t> pids;
-  for (int r = 0; r < size; ++r) {
-    pid_t pid = fork();
-    if (pid == 0) std::exit(worker(r, size, ndev, id, cfg, shared_out));
-    pids.push_back(pid);
-  }
-  int rc = 0;
-  for (pid_t pid : pids) {
-    int st = 0;
-    waitpid(pid, &st, 0);
-    if (!WIFEXITED(st) || WEXITSTATUS(st) != 0) rc = 1;
-  }
-  return rc;
+  hpk_launch::SharedBootstrap* sh = hpk_launch::map_shared();
+  return hpk_launch::fork_workers(
+      size, [&](int rank) { return worker(rank, size, ndev, sh, cfg); });
 }

@@ -23,6 +23,7 @@
 // reference's host sort+sum.
 
 #include "../hpc_patterns_amd/native/include/hpk.h"
+#include "launch_util.h"
 
 #include <rccl/rccl.h>
 #include <sys/socket.h>
@@ -124,6 +125,19 @@ int run_peer(size_t n, bool bidir_phase) {
     for (int it = 0; it < kIters; ++it) {
       double t0 = now_s();
       for (auto& pr : pairs) {
+        // same-device fallback must use a plain async copy:
+        // hipMemcpyPeerAsync with srcDevice==dstDevice returns success but
+        // moves nothing (observed on ROCm 7.2/gfx950).
+        if (pr.a == pr.b) {
+          hpk::check_hip(hipMemcpyAsync(pr.dst_b, pr.src_a, bytes,
+                                        hipMemcpyDeviceToDevice, pr.sa),
+                         "d2d");
+          if (bidir)
+            hpk::check_hip(hipMemcpyAsync(pr.dst_a, pr.src_b, bytes,
+                                          hipMemcpyDeviceToDevice, pr.sb),
+                           "d2d");
+          continue;
+        }
         hpk::memcpy_peer_async(pr.dst_b, pr.b, pr.src_a, pr.a, bytes, pr.sa);
         if (bidir)
           hpk::memcpy_peer_async(pr.dst_a, pr.a, pr.src_b, pr.b, bytes, pr.sb);
@@ -155,7 +169,8 @@ int run_peer(size_t n, bool bidir_phase) {
 // Engine 2: HIP-IPC one-sided put between two processes.
 // ---------------------------------------------------------------------------
 int run_ipc(size_t n) {
-  int ndev = hpk::device_count();
+  // IMPORTANT: fork BEFORE any HIP call — an initialized HIP runtime does
+  // not survive fork(), so each side initializes its own.
   size_t bytes = n * sizeof(float);
   int sv[2];
   if (socketpair(AF_UNIX, SOCK_STREAM, 0, sv) != 0) {
@@ -167,6 +182,7 @@ int run_ipc(size_t n) {
   if (pid == 0) {
     // child = rank 1 (the "origin" of the one-sided put)
     close(sv[0]);
+    int ndev = hpk::device_count();
     int dev = ndev >= 2 ? 1 : 0;
     hpk::check_hip(hipSetDevice(dev), "child set dev");
     // receive target handle
@@ -230,11 +246,12 @@ int run_ipc(size_t n) {
 // ---------------------------------------------------------------------------
 // Engine 3: RCCL pt2pt, one process per GPU.
 // ---------------------------------------------------------------------------
-int rccl_worker(int rank, int size, const ncclUniqueId& id, size_t n) {
+int rccl_worker(int rank, int size, hpk_launch::SharedBootstrap* sh, size_t n) {
   hpk::check_hip(hipSetDevice(rank), "set dev");
+  ncclUniqueId id;
+  hpk_launch::bootstrap_id(sh, rank, &id);
   ncclComm_t comm;
-  check_nccl(ncclCommInitRank(&comm, size, const_cast<ncclUniqueId&>(id), rank),
-             "init");
+  check_nccl(ncclCommInitRank(&comm, size, id, rank), "init");
   size_t bytes = n * sizeof(float);
   float *src = nullptr, *dst = nullptr;
   hpk::check_hip(hipMalloc(&src, bytes), "src");
@@ -305,31 +322,18 @@ int rccl_worker(int rank, int size, const ncclUniqueId& id, size_t n) {
   return 0;
 }
 
-int run_rccl(size_t n) {
-  int ndev = hpk::device_count();
+int run_rccl(const char* self, size_t n) {
+  // parent stays HIP-free (launch_util.h invariant)
+  int ndev = hpk_launch::probe_device_count(self);
+  hpk_launch::SharedBootstrap* sh = hpk_launch::map_shared();
   if (ndev < 2) {
     std::printf("# rccl engine needs >=2 GPUs (have %d) — running 1-rank "
                 "plumbing check\n", ndev);
-    ncclUniqueId id;
-    check_nccl(ncclGetUniqueId(&id), "id");
-    return rccl_worker(0, 1, id, n);
+    return rccl_worker(0, 1, sh, n);
   }
   int size = ndev - (ndev % 2); // even
-  ncclUniqueId id;
-  check_nccl(ncclGetUniqueId(&id), "id");
-  std::vector<pid_t> pids;
-  for (int r = 0; r < size; ++r) {
-    pid_t pid = fork();
-    if (pid == 0) std::exit(rccl_worker(r, size, id, n));
-    pids.push_back(pid);
-  }
-  int rc = 0;
-  for (pid_t p : pids) {
-    int st = 0;
-    waitpid(p, &st, 0);
-    if (!WIFEXITED(st) || WEXITSTATUS(st) != 0) rc = 1;
-  }
-  return rc;
+  return hpk_launch::fork_workers(
+      size, [&](int rank) { return rccl_worker(rank, size, sh, n); });
 }
 
 } // namespace
@@ -343,7 +347,10 @@ int main(int argc, char* argv[]) {
       if (++i >= argc) { std::fprintf(stderr, "missing value\n"); std::exit(1); }
       return argv[i];
     };
-    if (s == "--engine") engine = next();
+    if (s == "--probe-ndev") {  // launch_util.h re-exec probe
+      std::printf("%d\n", hpk::device_count());
+      return 0;
+    } else if (s == "--engine") engine = next();
     else if (s == "-n" || s == "--floats") n = std::strtoull(next(), nullptr, 10);
     else {
       std::printf("Usage: %s [--engine peer|ipc|rccl] [--floats N]\n", argv[0]);
@@ -352,7 +359,7 @@ int main(int argc, char* argv[]) {
   }
   if (engine == "peer") return run_peer(n, true);
   if (engine == "ipc") return run_ipc(n);
-  if (engine == "rccl") return run_rccl(n);
+  if (engine == "rccl") return run_rccl(argv[0], n);
   std::fprintf(stderr, "unknown engine '%s'\n", engine.c_str());
   return 1;
 }
