@@ -25,7 +25,7 @@ LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATI
 LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
 
 EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)
-BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p
+BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p $(BIN)/hpk_interop
 
 .PHONY: all ext bins clean
 all: ext bins
@@ -54,6 +54,9 @@ $(BIN)/hpk_allreduce: $(BUILD)/allreduce_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
 
 $(BIN)/hpk_p2p: $(BUILD)/p2p_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
+
+$(BIN)/hpk_interop: $(BUILD)/interop_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
 
 $(BUILD):

@@ -196,3 +196,9 @@ def test_bench_smoke_json():
                 "dtype", "data", "config"):
         assert key in data, key
     assert data["n_gpus"] == 1 and data["value"] > 0
+
+
+def test_bin_interop():
+    res = _run([str(REPO / "bin/hpk_interop")])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "PASSED" in res.stdout
