@@ -161,3 +161,21 @@ def body_flagship_cpu_accounting(rank, world):
 ])
 def test_dist_pattern(world, fn, dist_env):
     run_dist(world, fn, int(dist_env["MASTER_PORT"]))
+
+
+def body_sweep(rank, world):
+    import io
+    from contextlib import redirect_stdout
+
+    from hpc_patterns_amd.parallel import sweep
+
+    t = sweep.bench_algo("ring", 4096, iters=2,
+                         device=torch.device("cpu"))
+    assert t > 0
+    t2 = sweep.bench_algo("rsag", 4096, iters=2, device=torch.device("cpu"))
+    assert t2 > 0
+    return True
+
+
+def test_sweep_bench_algo(dist_env):
+    run_dist(2, "body_sweep", int(dist_env["MASTER_PORT"]))
