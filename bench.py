@@ -26,6 +26,48 @@ import sys
 import time
 
 
+class _CpuPlumbingStep:
+    """CI-only stub: exercises bench.py's exact distributed control flow
+    (collectives, pt2pt pairing, byte accounting) on gloo/CPU so the
+    multi-GPU path is correct by construction before it ever sees 8 GPUs."""
+
+    def __init__(self, rank, world, cfg):
+        import torch
+        import torch.distributed as dist
+
+        self.config = cfg
+        self.rank, self.world_size = rank, world
+        self.distributed = world > 1 and dist.is_initialized()
+        self.peer = (rank + 1 if rank % 2 == 0 else rank - 1)
+        if self.peer >= world:
+            self.peer = None
+        n = cfg["p2p_floats"]
+        self.ar_buf = torch.ones(cfg["allreduce_floats"])
+        self.p2p_send = torch.arange(n, dtype=torch.float32)
+        self.p2p_recv = torch.empty(n)
+        self.local = torch.empty(cfg["d2d_floats"])
+
+    def step(self):
+        import torch.distributed as dist
+
+        self.local.fill_(1.0)  # stand-in for the local stream bundle
+        if self.distributed:
+            work = dist.all_reduce(self.ar_buf, async_op=True)
+            if self.peer is not None:
+                reqs = dist.batch_isend_irecv([
+                    dist.P2POp(dist.isend, self.p2p_send, self.peer),
+                    dist.P2POp(dist.irecv, self.p2p_recv, self.peer),
+                ])
+                for r in reqs:
+                    r.wait()
+            work.wait()
+
+    def bytes_per_step_per_rank(self):
+        from hpc_patterns_amd.models.flagship import FlagshipPatternStep
+
+        return FlagshipPatternStep.bytes_per_step_per_rank(self)
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -33,11 +75,14 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--smoke", action="store_true",
                     help="tiny config (seconds, used by __graft_entry__)")
+    ap.add_argument("--cpu", action="store_true",
+                    help="CPU/gloo plumbing check of the distributed bench "
+                         "path (CI only — NOT a performance measurement)")
     args = ap.parse_args()
 
     import torch
 
-    if not torch.cuda.is_available():
+    if not torch.cuda.is_available() and not args.cpu:
         print("bench.py requires a GPU (MI355X)", file=sys.stderr)
         return 1
 
@@ -50,20 +95,29 @@ def main() -> int:
 
     world = int(os.environ.get("WORLD_SIZE", 1))
     if world > 1:
-        rank, local_rank, world = init_distributed()
+        rank, local_rank, world = init_distributed(
+            backend="gloo" if args.cpu else None)
     else:
         rank, local_rank = 0, 0
-        torch.cuda.set_device(0)
+        if not args.cpu:
+            torch.cuda.set_device(0)
 
-    device = torch.device("cuda", torch.cuda.current_device())
-    cfg = dict(SMOKE_CONFIG if args.smoke else DEFAULT_CONFIG)
-    step = FlagshipPatternStep(device=device, rank=rank, world_size=world,
-                               config=cfg)
+    if args.cpu:
+        device = torch.device("cpu")
+        cfg = dict(SMOKE_CONFIG)
+        cfg["tripcount"] = 1
+        step = _CpuPlumbingStep(rank, world, cfg)
+    else:
+        device = torch.device("cuda", torch.cuda.current_device())
+        cfg = dict(SMOKE_CONFIG if args.smoke else DEFAULT_CONFIG)
+        step = FlagshipPatternStep(device=device, rank=rank, world_size=world,
+                                   config=cfg)
 
     def barrier_sync():
         if world > 1:
             dist.barrier()
-        torch.cuda.synchronize()
+        if not args.cpu:
+            torch.cuda.synchronize()
 
     for _ in range(args.warmup):
         step.step()
@@ -87,12 +141,15 @@ def main() -> int:
 
     # ---- component diagnostics (outside the timed region) ----
     components = {}
-    overlap = step.measure_overlap(reps=3)
-    components["stream_overlap_pct"] = round(
-        100.0 * overlap["overlap_efficiency"], 1)
-    components["stream_overlap_speedup"] = round(overlap["speedup"], 3)
-    components["theoretical_speedup"] = round(
-        overlap["theoretical_speedup"], 3)
+    if not args.cpu:
+        overlap = step.measure_overlap(reps=3)
+        components["stream_overlap_pct"] = round(
+            100.0 * overlap["overlap_efficiency"], 1)
+        components["stream_overlap_speedup"] = round(overlap["speedup"], 3)
+        components["theoretical_speedup"] = round(
+            overlap["theoretical_speedup"], 3)
+        components["per_command_ms"] = [
+            round(t * 1e3, 3) for t in overlap["per_command_s"]]
     if world > 1:
         bw = pairwise_bandwidth(cfg["p2p_floats"] * 4, iters=5,
                                 bidirectional=False, device=device)
@@ -103,7 +160,8 @@ def main() -> int:
 
     if rank == 0:
         result = {
-            "metric": "pattern_aggregate_GBps",
+            "metric": ("cpu_plumbing_check" if args.cpu
+                       else "pattern_aggregate_GBps"),
             "value": round(agg_gbps, 2),
             "unit": "GB/s",
             "n_gpus": world,

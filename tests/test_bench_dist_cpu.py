@@ -1,0 +1,49 @@
+"""End-to-end CPU check of the bench.py driver contract under torchrun:
+2 ranks, gloo, exactly the launch line the driver uses for N>1 (with --cpu).
+Validates rendezvous, the timed-region bracketing, MAX-over-ranks, and the
+one-JSON-line output."""
+
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+
+pytestmark = pytest.mark.timeout(300)
+
+
+def test_bench_torchrun_cpu(dist_env):
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        str(REPO / "bench.py"), "--gpus", "2", "--steps", "3",
+        "--warmup", "1", "--cpu",
+    ]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    json_lines = [l for l in res.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, res.stdout  # exactly one line, from rank 0
+    data = json.loads(json_lines[0])
+    assert data["metric"] == "cpu_plumbing_check"
+    assert data["n_gpus"] == 2
+    assert data["steps"] == 3
+    assert data["config"]["parallelism"] == "dp2"
+    assert "pingpong_us" in data["components"]
+    assert data["components"]["p2p_checksum_ok"] is True
+
+
+def test_bench_single_rank_cpu():
+    res = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "2",
+         "--warmup", "1", "--cpu"],
+        capture_output=True, text=True, timeout=120, cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    data = json.loads([l for l in res.stdout.splitlines()
+                       if l.startswith("{")][-1])
+    assert data["n_gpus"] == 1 and data["value"] > 0
