@@ -63,6 +63,16 @@ PYBIND11_MODULE(_hpk, m) {
                                   as_stream(stream));
         },
         py::arg("dst"), py::arg("src"), py::arg("nbytes"), py::arg("stream") = 0);
+  m.def("copy_kernel_tuned",
+        [](uintptr_t dst, uintptr_t src, size_t nbytes, uintptr_t stream,
+           int unroll, size_t grid_cap) {
+          hpk::launch_copy_kernel_tuned(reinterpret_cast<void*>(dst),
+                                        reinterpret_cast<const void*>(src),
+                                        nbytes, as_stream(stream), unroll,
+                                        grid_cap);
+        },
+        py::arg("dst"), py::arg("src"), py::arg("nbytes"), py::arg("stream") = 0,
+        py::arg("unroll") = 1, py::arg("grid_cap") = 16384);
   m.def("fill_f32",
         [](uintptr_t dst, float value, size_t n, uintptr_t stream) {
           hpk::launch_fill_f32(reinterpret_cast<float*>(dst), value, n,

@@ -44,6 +44,11 @@ void launch_busy_wait_mfma(float* out, long tripcount, long n_waves,
 // of hipMemcpyAsync's SDMA path). Pointers may be in any HIP-visible space.
 void launch_copy_kernel(void* dst, const void* src, size_t nbytes,
                         hipStream_t stream);
+// Tunable variant for micro-benchmarking: unroll in {1,4} (16 B vs 64 B per
+// thread per iteration), grid_cap = max workgroups.
+void launch_copy_kernel_tuned(void* dst, const void* src, size_t nbytes,
+                              hipStream_t stream, int unroll,
+                              size_t grid_cap);
 
 // K4 fills (reference Initialize kernel, allreduce-mpi-sycl.cpp:34-41).
 void launch_fill_f32(float* dst, float value, size_t n, hipStream_t stream);
