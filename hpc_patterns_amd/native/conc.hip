@@ -23,6 +23,7 @@
 #include <algorithm>
 #include <chrono>
 #include <cstring>
+#include <cstdlib>
 #include <limits>
 #include <numeric>
 #include <stdexcept>
@@ -62,10 +63,18 @@ struct Buffer {
       case 'D':
         check_hip(hipMalloc(&ptr, nbytes), "hipMalloc");
         break;
-      case 'H':
-        check_hip(hipHostMalloc(&ptr, nbytes, hipHostMallocDefault),
-                  "hipHostMalloc");
+      case 'H': {
+        // HPK_PINNED_FLAGS sweeps the pinned-allocation flavour (the ROCm
+        // analog of the reference's copy-engine env knobs): default | nc
+        // (non-coherent, cacheable) | wc (write-combined).
+        unsigned flags = hipHostMallocDefault;
+        const char* env = std::getenv("HPK_PINNED_FLAGS");
+        if (env && std::string(env) == "nc") flags = hipHostMallocNonCoherent;
+        if (env && std::string(env) == "wc")
+          flags = hipHostMallocWriteCombined | hipHostMallocMapped;
+        check_hip(hipHostMalloc(&ptr, nbytes, flags), "hipHostMalloc");
         break;
+      }
       case 'S':
         check_hip(hipMallocManaged(&ptr, nbytes, hipMemAttachGlobal),
                   "hipMallocManaged");
@@ -221,6 +230,7 @@ ConcResult conc_bench(const std::string& mode,
   }
 
   // ---- measured repetitions ----
+  trace_push(("conc_bench:" + mode).c_str());
   for (int rep = 0; rep < n_repetitions; ++rep) {
     long t0 = now_us();
     if (serial) {
@@ -276,6 +286,7 @@ ConcResult conc_bench(const std::string& mode,
     if (verbose)
       fprintf(stderr, "# rep %d: %ld us\n", rep, now_us() - t0);
   }
+  trace_pop();
 
   if (serial) {
     // Floor the serial total by the sum of per-command minima — the tightest

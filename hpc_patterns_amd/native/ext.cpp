@@ -155,6 +155,9 @@ PYBIND11_MODULE(_hpk, m) {
     hpk::ipc_close_handle(reinterpret_cast<void*>(dptr));
   });
   m.def("enable_peer_access", &hpk::enable_peer_access);
+  m.def("trace_push", [](const std::string& n) { hpk::trace_push(n.c_str()); });
+  m.def("trace_pop", &hpk::trace_pop);
+  m.def("trace_mark", [](const std::string& n) { hpk::trace_mark(n.c_str()); });
   m.def("memcpy_peer_async",
         [](uintptr_t dst, int dst_dev, uintptr_t src, int src_dev,
            size_t nbytes, uintptr_t stream) {

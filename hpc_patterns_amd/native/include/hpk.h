@@ -131,6 +131,13 @@ std::vector<uint8_t> ipc_get_handle(void* dptr);
 void* ipc_open_handle(const std::vector<uint8_t>& handle);
 void ipc_close_handle(void* dptr);
 
+// ---------------------------------------------------------------------------
+// Tracing (trace.hip) — roctx ranges for rocprofv3 --marker-trace.
+// ---------------------------------------------------------------------------
+void trace_push(const char* name);
+void trace_pop();
+void trace_mark(const char* name);
+
 void enable_peer_access(int peer_device);
 void memcpy_peer_async(void* dst, int dst_dev, const void* src, int src_dev,
                        size_t nbytes, hipStream_t stream);

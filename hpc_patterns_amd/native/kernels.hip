@@ -26,10 +26,11 @@ void check_hip(hipError_t e, const char* what) {
 namespace {
 
 constexpr int kBlock = 256; // 4 wave64 per workgroup
-// Streaming grid cap: 256 CUs want >> 256 workgroups in flight; 16384 blocks
-// of 256 threads = 4.2M threads, plenty for 8 XCDs without oversubscribing
-// the dispatcher.
-constexpr size_t kMaxGrid = 16384;
+// Streaming grid cap: 256 CUs want >> 256 workgroups in flight. Measured on
+// MI355X (profiles/membench_1gpu_r3.log): 1 GiB copy at cap 65536 runs
+// 2.60 TB/s payload vs 2.32 at 16384 — fewer grid-stride iterations per
+// thread wins for pure streaming.
+constexpr size_t kMaxGrid = 65536;
 
 inline size_t stream_grid(size_t n_items) {
   size_t blocks = (n_items + kBlock - 1) / kBlock;
@@ -76,13 +77,18 @@ __global__ void k_busy_wait_mfma(float* __restrict__ out, long tripcount) {
   short seed = (short)(threadIdx.x + 1);
   bf16x8_t a = {seed, seed, seed, seed, seed, seed, seed, seed};
   bf16x8_t b = {1, 2, 3, 4, 5, 6, 7, 8};
-  f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
-  for (long t = 0; t < tripcount; ++t) {
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  // Two independent accumulators: the 16x16x32 MFMA's dependent-accumulator
+  // latency exceeds its issue interval, so a single chain leaves the pipe
+  // under-issued (measured 1.63 PF with one chain at 2 waves/SIMD).
+  f32x4_t acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t acc1 = {0.f, 0.f, 0.f, 0.f};
+  for (long t = 0; t < tripcount; t += 2) {
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc1, 0, 0, 0);
   }
-  // One store per thread keeps the accumulator alive without noise.
+  // One store per thread keeps the accumulators alive without noise.
   out[(size_t)blockIdx.x * blockDim.x + threadIdx.x] =
-      acc[0] * 1e-30f; // scaled so huge accumulations don't overflow readers
+      (acc0[0] + acc1[0]) * 1e-30f; // scaled to avoid overflow for readers
 }
 
 // --------------------------------------------------------------------------
@@ -178,10 +184,30 @@ __global__ void k_acc_f1(float* __restrict__ dst, const float* __restrict__ src,
 // --------------------------------------------------------------------------
 __global__ void k_reduce_partial_f32(const float* __restrict__ src, size_t n,
                                      double* __restrict__ partial) {
-  double s = 0.0;
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  // float4 loads + 4 independent f64 accumulators: breaks the serial f64-add
+  // dependency chain that capped the first version at 2.2 TB/s read.
+  double s0 = 0.0, s1 = 0.0, s2 = 0.0, s3 = 0.0;
   size_t stride = (size_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) s += (double)src[i];
+  size_t n4 = n / 4;
+  const float4* src4 = (const float4*)src;
+  if (((uintptr_t)src % 16) == 0) {
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+         i += stride) {
+      float4 v = src4[i];
+      s0 += (double)v.x;
+      s1 += (double)v.y;
+      s2 += (double)v.z;
+      s3 += (double)v.w;
+    }
+    // tail elements by block 0 / thread 0..3
+    if (blockIdx.x == 0 && threadIdx.x < n - n4 * 4)
+      s0 += (double)src[n4 * 4 + threadIdx.x];
+  } else {
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+      s0 += (double)src[i];
+  }
+  double s = (s0 + s1) + (s2 + s3);
 
   // wave64 reduction
   for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
@@ -309,7 +335,7 @@ void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream) 
 }
 
 double reduce_sum_f32(const float* src, size_t n, hipStream_t stream) {
-  constexpr size_t kRedBlocks = 1024;
+  constexpr size_t kRedBlocks = 4096;
   size_t blocks = (n + (size_t)kBlock * 8 - 1) / ((size_t)kBlock * 8);
   if (blocks == 0) blocks = 1;
   if (blocks > kRedBlocks) blocks = kRedBlocks;
