@@ -64,5 +64,19 @@ $(BUILD):
 $(BIN):
 	mkdir -p $(BIN)
 
+# Race/memory-error hunting build: host-side ASan on the
+# binaries. Run on a GPU box with ASAN_OPTIONS=detect_leaks=0 (the HIP
+# runtime intentionally holds allocations).
+asan: CXXFLAGS += -fsanitize=address -g1
+asan: clean-bins $(BINARIES)
+
+clean-bins:
+	rm -rf $(BIN) $(filter %_main.o,$(wildcard $(BUILD)/*.o))
+
+test:
+	$(PYTHON) -m pytest tests -q -m "not gpu"
+
 clean:
 	rm -rf $(BUILD) $(BIN) $(EXT_SO)
+
+.PHONY: asan clean-bins test
