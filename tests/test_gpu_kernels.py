@@ -129,3 +129,27 @@ def test_interop_torch_allocator_shared_with_hip_kernels(ops, dev):
         u = t * 2.0                        # torch op, same stream, same memory
     s.synchronize()
     assert torch.equal(u, torch.full_like(t, 14.0))
+
+
+@pytest.mark.parametrize("unroll,cap", [(1, 4096), (4, 65536)])
+def test_copy_kernel_tuned(ops, dev, unroll, cap):
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    n = (1 << 20) + 5
+    src = torch.rand(n, device=dev)
+    dst = torch.zeros_like(src)
+    torch.cuda.synchronize()
+    hpk.copy_kernel_tuned(dst.data_ptr(), src.data_ptr(), n * 4,
+                          torch.cuda.current_stream().cuda_stream, unroll, cap)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, src)
+
+
+def test_roctx_markers_noop_safe(ops, dev):
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    hpk.trace_push("test_range")
+    hpk.trace_mark("test_mark")
+    hpk.trace_pop()
