@@ -34,6 +34,7 @@
 //   --algo ring|pipeline|rccl
 
 #include "../hpc_patterns_amd/native/include/hpk.h"
+#include "../hpc_patterns_amd/native/include/rccl_datatype.h"
 #include "launch_util.h"
 
 #include <rccl/rccl.h>
@@ -70,9 +71,35 @@ struct Config {
   int p = 25;
   char alloc = 'D';
   std::string algo = "ring";
+  std::string dtype = "float"; // float | int (reference -DAPP_DATA_TYPE pair)
   int nranks = -1;
   int iters = 10;
   int chunks = 8;
+};
+
+// dtype dispatch for the device kernels (fill / accumulate / exact sum)
+template <typename T> struct Kern;
+template <> struct Kern<float> {
+  static void fill(float* p, float v, size_t n, hipStream_t s) {
+    hpk::launch_fill_f32(p, v, n, s);
+  }
+  static void acc(float* d, const float* s_, size_t n, hipStream_t s) {
+    hpk::launch_acc_f32(d, s_, n, s);
+  }
+  static double sum(const float* p, size_t n, hipStream_t s) {
+    return hpk::reduce_sum_f32(p, n, s);
+  }
+};
+template <> struct Kern<int> {
+  static void fill(int* p, int v, size_t n, hipStream_t s) {
+    hpk::launch_fill_i32(p, v, n, s);
+  }
+  static void acc(int* d, const int* s_, size_t n, hipStream_t s) {
+    hpk::launch_acc_i32(d, s_, n, s);
+  }
+  static double sum(const int* p, size_t n, hipStream_t s) {
+    return (double)hpk::reduce_sum_i32(p, n, s);
+  }
 };
 
 void* alloc_buf(char kind, size_t bytes) {
@@ -95,21 +122,23 @@ void* alloc_buf(char kind, size_t bytes) {
 // Hand ring all-reduce, the reference SendRecvRing pattern: (size-1) steps of
 // [exchange full buffer with ring neighbours] + [VC += recv]. Buffers VA
 // (send payload, swapped with VB each step), VB (recv), VC (accumulator).
-double run_ring(ncclComm_t comm, hipStream_t stream, float* va, float* vb,
-                float* vc, size_t n, int rank, int size) {
+template <typename T>
+double run_ring(ncclComm_t comm, hipStream_t stream, T* va, T* vb,
+                T* vc, size_t n, int rank, int size) {
+  const ncclDataType_t dt = hpk::get_rccl_datatype<T>();
   int right = (rank + 1) % size;
   int left = (rank - 1 + size) % size;
   double t0 = now_s();
-  hpk::launch_acc_f32(vc, va, n, stream); // VC += own VA (VC starts at 0)
+  Kern<T>::acc(vc, va, n, stream); // VC += own VA (VC starts at 0)
   for (int step = 0; step < size - 1; ++step) {
     // RCCL pt2pt: group makes the send+recv concurrent (no odd/even ordering
     // dance needed — that deadlock-avoidance trick is an MPI-blocking-call
     // artifact, reference allreduce-mpi-sycl.cpp:50-58).
     check_nccl(ncclGroupStart(), "group start");
-    check_nccl(ncclSend(va, n, ncclFloat, right, comm, stream), "send");
-    check_nccl(ncclRecv(vb, n, ncclFloat, left, comm, stream), "recv");
+    check_nccl(ncclSend(va, n, dt, right, comm, stream), "send");
+    check_nccl(ncclRecv(vb, n, dt, left, comm, stream), "recv");
     check_nccl(ncclGroupEnd(), "group end");
-    hpk::launch_acc_f32(vc, vb, n, stream);
+    Kern<T>::acc(vc, vb, n, stream);
     std::swap(va, vb);
   }
   hpk::check_hip(hipStreamSynchronize(stream), "ring sync");
@@ -119,9 +148,11 @@ double run_ring(ncclComm_t comm, hipStream_t stream, float* va, float* vb,
 // Chunked pipelined ring: split the buffer into K chunks; while chunk c is
 // being accumulated on the compute stream, chunk c+1 is already in flight on
 // the comm stream. Overlaps xGMI transfer with the HIP Accumulate kernel.
+template <typename T>
 double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
-                    hipStream_t comp_stream, float* va, float* vb, float* vc,
+                    hipStream_t comp_stream, T* va, T* vb, T* vc,
                     size_t n, int rank, int size, int chunks) {
+  const ncclDataType_t dt = hpk::get_rccl_datatype<T>();
   int right = (rank + 1) % size;
   int left = (rank - 1 + size) % size;
   size_t chunk = (n + chunks - 1) / chunks;
@@ -130,7 +161,7 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
     hpk::check_hip(hipEventCreateWithFlags(&e, hipEventDisableTiming), "ev");
 
   double t0 = now_s();
-  hpk::launch_acc_f32(vc, va, n, comp_stream);
+  Kern<T>::acc(vc, va, n, comp_stream);
   for (int step = 0; step < size - 1; ++step) {
     // launch all chunk exchanges; accumulate each chunk as soon as it lands
     for (int c = 0; c < chunks; ++c) {
@@ -138,14 +169,14 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
       if (off >= n) break;
       size_t len = std::min(chunk, n - off);
       check_nccl(ncclGroupStart(), "group start");
-      check_nccl(ncclSend(va + off, len, ncclFloat, right, comm, comm_stream),
+      check_nccl(ncclSend(va + off, len, dt, right, comm, comm_stream),
                  "send");
-      check_nccl(ncclRecv(vb + off, len, ncclFloat, left, comm, comm_stream),
+      check_nccl(ncclRecv(vb + off, len, dt, left, comm, comm_stream),
                  "recv");
       check_nccl(ncclGroupEnd(), "group end");
       hpk::check_hip(hipEventRecord(done[c], comm_stream), "record");
       hpk::check_hip(hipStreamWaitEvent(comp_stream, done[c], 0), "wait");
-      hpk::launch_acc_f32(vc + off, vb + off, len, comp_stream);
+      Kern<T>::acc(vc + off, vb + off, len, comp_stream);
     }
     // comm of next step must not overwrite vb before accumulate read it:
     // swap uses distinct buffers, but step s+1 recv into (old) va after
@@ -162,6 +193,7 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
   return dt;
 }
 
+template <typename T>
 int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
            const Config& cfg) {
   int dev = rank % ndev;
@@ -172,10 +204,10 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
   check_nccl(ncclCommInitRank(&comm, size, id, rank), "ncclCommInitRank");
 
   size_t n = 1ull << cfg.p;
-  size_t bytes = n * sizeof(float);
-  float* va = (float*)alloc_buf(cfg.alloc, bytes);
-  float* vb = (float*)alloc_buf(cfg.alloc, bytes);
-  float* vc = (float*)alloc_buf(cfg.alloc, bytes);
+  size_t bytes = n * sizeof(T);
+  T* va = (T*)alloc_buf(cfg.alloc, bytes);
+  T* vb = (T*)alloc_buf(cfg.alloc, bytes);
+  T* vc = (T*)alloc_buf(cfg.alloc, bytes);
 
   hipStream_t stream, comp_stream;
   hpk::check_hip(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking), "s");
@@ -185,15 +217,16 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
   double best = 1e30;
   for (int it = 0; it < cfg.iters; ++it) {
     // (re)initialize: VA = rank, VB = -1, VC = 0 (reference Initialize)
-    hpk::launch_fill_f32(va, (float)rank, n, stream);
-    hpk::launch_fill_f32(vb, -1.f, n, stream);
-    hpk::launch_fill_f32(vc, 0.f, n, stream);
+    Kern<T>::fill(va, (T)rank, n, stream);
+    Kern<T>::fill(vb, (T)-1, n, stream);
+    Kern<T>::fill(vc, (T)0, n, stream);
     hpk::check_hip(hipStreamSynchronize(stream), "init sync");
 
     double dt;
     if (cfg.algo == "rccl") {
       double t0 = now_s();
-      check_nccl(ncclAllReduce(va, vc, n, ncclFloat, ncclSum, comm, stream),
+      check_nccl(ncclAllReduce(va, vc, n, hpk::get_rccl_datatype<T>(),
+                               ncclSum, comm, stream),
                  "ncclAllReduce");
       hpk::check_hip(hipStreamSynchronize(stream), "allreduce sync");
       dt = now_s() - t0;
@@ -221,7 +254,7 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
 
   // analytic verification: every element == size*(size-1)/2
   double expected = (double)n * ((double)size * (size - 1) / 2.0);
-  double got = hpk::reduce_sum_f32(vc, n, stream);
+  double got = Kern<T>::sum(vc, n, stream);
   bool pass = std::abs(got - expected) < 1e-6 * std::max(1.0, expected);
   std::printf("%s rank %d (sum %.1f, expected %.1f)\n",
               pass ? "Passed" : "FAILED", rank, got, expected);
@@ -231,9 +264,10 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
     // bus bandwidth convention: ring moves 2(size-1)/size * bytes per rank
     double busbw =
         size > 1 ? 2.0 * (size - 1) / size * gb / max_time : gb / max_time;
-    std::printf("# algo=%s ranks=%d elems=2^%d alloc=%c time=%.6fs "
-                "busbw=%.2f GB/s\n",
-                cfg.algo.c_str(), size, cfg.p, cfg.alloc, max_time, busbw);
+    std::printf("# algo=%s dtype=%s ranks=%d elems=2^%d alloc=%c "
+                "time=%.6fs busbw=%.2f GB/s\n",
+                cfg.algo.c_str(), cfg.dtype.c_str(), size, cfg.p, cfg.alloc,
+                max_time, busbw);
     if (sh) sh->result = max_time;
   }
 
@@ -264,6 +298,7 @@ int main(int argc, char* argv[]) {
     else if (s == "-i") cfg.iters = std::atoi(next());
     else if (s == "-c") cfg.chunks = std::atoi(next());
     else if (s == "--algo") cfg.algo = next();
+    else if (s == "-t" || s == "--dtype") cfg.dtype = next();
     else {
       std::printf(
           "Usage: %s [-p P] [-D|-H|-S] [-a] [-n ranks] [-i iters] [-c chunks] "
@@ -294,6 +329,8 @@ int main(int argc, char* argv[]) {
   }
 
   hpk_launch::SharedBootstrap* sh = hpk_launch::map_shared();
-  return hpk_launch::fork_workers(
-      size, [&](int rank) { return worker(rank, size, ndev, sh, cfg); });
+  return hpk_launch::fork_workers(size, [&](int rank) {
+    return cfg.dtype == "int" ? worker<int>(rank, size, ndev, sh, cfg)
+                              : worker<float>(rank, size, ndev, sh, cfg);
+  });
 }

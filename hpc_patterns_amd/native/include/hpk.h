@@ -66,6 +66,11 @@ void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream);
 // (peer2pear.cpp:56-63) with an order-independent exact device reduction.
 double reduce_sum_f32(const float* src, size_t n, hipStream_t stream);
 
+// int32 twins (the reference instantiates miniapps for float AND int).
+void launch_fill_i32(int* dst, int value, size_t n, hipStream_t stream);
+void launch_acc_i32(int* dst, const int* src, size_t n, hipStream_t stream);
+long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream);
+
 // ---------------------------------------------------------------------------
 // Concurrency engine (conc.hip) — reference bench<T>() ABI, bench.hpp:37-40.
 // ---------------------------------------------------------------------------

@@ -178,6 +178,65 @@ __global__ void k_acc_f1(float* __restrict__ dst, const float* __restrict__ src,
 }
 
 // --------------------------------------------------------------------------
+// int32 twins of fill/accumulate (the reference instantiates its miniapps
+// for float AND int via -DAPP_DATA_TYPE, mpi-sycl/CMakeLists.txt:4-5).
+// --------------------------------------------------------------------------
+__global__ void k_fill_i4(int4* __restrict__ dst, int v, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  int4 val = make_int4(v, v, v, v);
+  for (; i < n4; i += stride) dst[i] = val;
+}
+
+__global__ void k_fill_i1(int* __restrict__ dst, int v, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = v;
+}
+
+__global__ void k_acc_i4(int4* __restrict__ dst, const int4* __restrict__ src,
+                         size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    int4 d = dst[i];
+    int4 s = src[i];
+    d.x += s.x;
+    d.y += s.y;
+    d.z += s.z;
+    d.w += s.w;
+    dst[i] = d;
+  }
+}
+
+__global__ void k_acc_i1(int* __restrict__ dst, const int* __restrict__ src,
+                         size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] += src[i];
+}
+
+__global__ void k_reduce_partial_i32(const int* __restrict__ src, size_t n,
+                                     long long* __restrict__ partial) {
+  long long s = 0;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    s += src[i];
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  __shared__ long long wsum_i[kBlock / 64];
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  if (lane == 0) wsum_i[wave] = s;
+  __syncthreads();
+  if (wave == 0) {
+    long long b = (lane < kBlock / 64) ? wsum_i[lane] : 0;
+    for (int off = 32; off > 0; off >>= 1) b += __shfl_down(b, off, 64);
+    if (lane == 0) partial[blockIdx.x] = b;
+  }
+}
+
+// --------------------------------------------------------------------------
 // Exact checksum: double partial sums per block (wave64 shuffle reduction,
 // then LDS across the block's 4 waves), host-side final sum of <=1024
 // partials.
@@ -332,6 +391,61 @@ void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream) 
                        dst, src, n);
   }
   check_hip(hipGetLastError(), "launch_acc_f32");
+}
+
+void launch_fill_i32(int* dst, int value, size_t n, hipStream_t stream) {
+  if (((uintptr_t)dst % 16 == 0) && n >= 4) {
+    size_t n4 = n / 4;
+    size_t tail = n - n4 * 4;
+    hipLaunchKernelGGL(k_fill_i4, dim3(stream_grid(n4)), dim3(kBlock), 0,
+                       stream, (int4*)dst, value, n4);
+    if (tail)
+      hipLaunchKernelGGL(k_fill_i1, dim3(1), dim3(64), 0, stream, dst + n4 * 4,
+                         value, tail);
+  } else {
+    hipLaunchKernelGGL(k_fill_i1, dim3(stream_grid(n)), dim3(kBlock), 0,
+                       stream, dst, value, n);
+  }
+  check_hip(hipGetLastError(), "launch_fill_i32");
+}
+
+void launch_acc_i32(int* dst, const int* src, size_t n, hipStream_t stream) {
+  if (((uintptr_t)dst % 16 == 0) && ((uintptr_t)src % 16 == 0) && n >= 4) {
+    size_t n4 = n / 4;
+    size_t tail = n - n4 * 4;
+    hipLaunchKernelGGL(k_acc_i4, dim3(stream_grid(n4)), dim3(kBlock), 0,
+                       stream, (int4*)dst, (const int4*)src, n4);
+    if (tail)
+      hipLaunchKernelGGL(k_acc_i1, dim3(1), dim3(64), 0, stream, dst + n4 * 4,
+                         src + n4 * 4, tail);
+  } else {
+    hipLaunchKernelGGL(k_acc_i1, dim3(stream_grid(n)), dim3(kBlock), 0, stream,
+                       dst, src, n);
+  }
+  check_hip(hipGetLastError(), "launch_acc_i32");
+}
+
+long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream) {
+  constexpr size_t kRedBlocks = 4096;
+  size_t blocks = (n + (size_t)kBlock * 8 - 1) / ((size_t)kBlock * 8);
+  if (blocks == 0) blocks = 1;
+  if (blocks > kRedBlocks) blocks = kRedBlocks;
+  long long* d_partial = nullptr;
+  check_hip(hipMallocAsync((void**)&d_partial, blocks * sizeof(long long),
+                           stream),
+            "reduce_sum_i32 hipMallocAsync");
+  hipLaunchKernelGGL(k_reduce_partial_i32, dim3(blocks), dim3(kBlock), 0,
+                     stream, src, n, d_partial);
+  check_hip(hipGetLastError(), "reduce_sum_i32 kernel");
+  std::vector<long long> h(blocks);
+  check_hip(hipMemcpyAsync(h.data(), d_partial, blocks * sizeof(long long),
+                           hipMemcpyDeviceToHost, stream),
+            "reduce_sum_i32 D2H");
+  check_hip(hipFreeAsync(d_partial, stream), "reduce_sum_i32 hipFreeAsync");
+  check_hip(hipStreamSynchronize(stream), "reduce_sum_i32 sync");
+  long long s = 0;
+  for (long long v : h) s += v;
+  return s;
 }
 
 double reduce_sum_f32(const float* src, size_t n, hipStream_t stream) {

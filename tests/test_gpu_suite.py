@@ -202,3 +202,10 @@ def test_bin_interop():
     res = _run([str(REPO / "bin/hpk_interop")])
     assert res.returncode == 0, res.stdout + res.stderr
     assert "PASSED" in res.stdout
+
+
+def test_bin_allreduce_int_dtype():
+    res = _run([str(REPO / "bin/hpk_allreduce"), "-p", "18", "-i", "1",
+                "-n", "1", "--algo", "rccl", "-t", "int"])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "Passed rank 0" in res.stdout and "dtype=int" in res.stdout
