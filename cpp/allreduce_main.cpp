@@ -188,9 +188,9 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
   }
   hpk::check_hip(hipStreamSynchronize(comp_stream), "pipeline sync");
   hpk::check_hip(hipStreamSynchronize(comm_stream), "pipeline sync comm");
-  double dt = now_s() - t0;
+  double elapsed = now_s() - t0;
   for (auto& e : done) (void)hipEventDestroy(e);
-  return dt;
+  return elapsed;
 }
 
 template <typename T>
