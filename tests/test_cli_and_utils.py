@@ -113,3 +113,17 @@ def test_gpu_mapping_script_policies(tmp_path):
                          capture_output=True, text=True, env=env)
     assert res.returncode == 0, res.stderr
     assert "dev=4" in res.stdout
+
+
+def test_env_knobs_listing(monkeypatch):
+    from hpc_patterns_amd.utils.config import check_env, env_knobs
+
+    knobs = env_knobs()
+    assert "HSA_ENABLE_SDMA" in knobs and "GPU_MAX_HW_QUEUES" in knobs
+    monkeypatch.setenv("HSA_ENABLE_SDMA", "0")
+    monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "1")
+    warns = check_env()
+    assert len(warns) == 2
+    monkeypatch.setenv("HSA_ENABLE_SDMA", "1")
+    monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    assert check_env() == []
