@@ -26,6 +26,45 @@ import sys
 import time
 
 
+def _measure_xgmi_peer_copy(nbytes: int, iters: int = 5):
+    """Direct hipMemcpyPeerAsync GB/s between GPU 0 and 1 (single process
+    sees the whole node under torchrun) — the xGMI SDMA path, complementing
+    the RCCL pt2pt number. Returns None when <2 GPUs are visible."""
+    import time
+
+    import torch
+
+    from hpc_patterns_amd._native import native
+
+    if torch.cuda.device_count() < 2:
+        return None
+    hpk = native()
+    cur = torch.cuda.current_device()
+    try:
+        hpk.set_device(0)
+        hpk.enable_peer_access(1)
+        src = hpk.hip_malloc(nbytes)
+        hpk.set_device(1)
+        hpk.enable_peer_access(0)
+        dst = hpk.hip_malloc(nbytes)
+        hpk.set_device(0)
+        best = float("inf")
+        for _ in range(iters + 1):  # first iter = warmup
+            t0 = time.perf_counter()
+            hpk.memcpy_peer_async(dst, 1, src, 0, nbytes, 0)
+            hpk.stream_synchronize(0)
+            dt = time.perf_counter() - t0
+            best = min(best, dt)
+        hpk.hip_free(src)
+        hpk.set_device(1)
+        hpk.hip_free(dst)
+        return nbytes / best / 1e9
+    except Exception:
+        return None
+    finally:
+        torch.cuda.set_device(cur)
+
+
 class _CpuPlumbingStep:
     """CI-only stub: exercises bench.py's exact distributed control flow
     (collectives, pt2pt pairing, byte accounting) on gloo/CPU so the
@@ -157,6 +196,12 @@ def main() -> int:
         components["p2p_checksum_ok"] = bool(bw["checksum_ok"])
         pp = pingpong(nbytes=8, iters=50, device=device)
         components["pingpong_us"] = round(pp["oneway_us"], 2)
+        if rank == 0 and not args.cpu:
+            xgmi = _measure_xgmi_peer_copy(cfg["p2p_floats"] * 4)
+            if xgmi is not None:
+                components["xgmi_peer_copy_GBps"] = round(xgmi, 2)
+        if world > 1:
+            dist.barrier()
 
     if rank == 0:
         result = {
