@@ -164,24 +164,56 @@ ConcResult conc_bench(const std::string& mode,
   }
 
   // ---- streams & events ----
-  // Graph mode needs one extra stream: streams[0] is the capture master and
-  // each command branches onto its own capture stream.
-  //
-  // Streams come from a process-lifetime pool: ROCm assigns SDMA engines to
-  // streams in creation order, so creating fresh streams per call makes the
-  // H2D/D2H engine pairing depend on call history (measured: the same
-  // H2D||D2H list overlapped 1.7x when run first but 0.9x when run after
-  // another list in the same process). The pool pins command i to the same
-  // stream/engine every call.
-  static std::vector<hipStream_t> pool;
-  size_t need = graph_mode ? (size_t)n_queues + 1 : (size_t)std::max(n_queues, 1);
-  while (pool.size() < need) {
-    hipStream_t s;
-    check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking),
-              "stream create");
-    pool.push_back(s);
+  // TYPED process-lifetime stream pools. Measured on MI355X (ROCm 7.2,
+  // profiles/pycli_r7 vs hd_dh_sizes_r6): once a stream has ever run a
+  // KERNEL, the runtime routes that stream's pinned H2D/D2H copies through
+  // the shader-blit path instead of SDMA — serial bandwidth halves
+  // (57 -> 29 GB/s) and H2D||D2H stops overlapping. So streams are drawn
+  // from three pools by command type — compute kernels, host<->device
+  // copies, device<->device copies — and copy streams never see a kernel.
+  static std::vector<hipStream_t> pool_kernel, pool_hostcopy, pool_devcopy;
+  static hipStream_t master_stream = nullptr;
+  auto take = [&](std::vector<hipStream_t>& pool, size_t idx) {
+    while (pool.size() <= idx) {
+      hipStream_t s;
+      check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking),
+                "stream create");
+      pool.push_back(s);
+    }
+    return pool[idx];
+  };
+  if (master_stream == nullptr)
+    check_hip(hipStreamCreateWithFlags(&master_stream, hipStreamNonBlocking),
+              "master stream create");
+
+  // stream for command i, honouring n_queues per type (round-robin inside a
+  // type; distinct types never share a stream).
+  std::vector<hipStream_t> cmd_stream(ncmds);
+  {
+    size_t nk = 0, nh = 0, nd = 0;
+    for (int i = 0; i < ncmds; ++i) {
+      const Command& c = cmds[i];
+      // a --copy_kernel shader copy SUBMITS a kernel -> kernel pool, so it
+      // cannot taint a copy stream
+      bool submits_kernel =
+          c.is_compute ||
+          (c.shader_copy && c.src.space != 'M' && c.dst.space != 'M');
+      bool host_side = !c.is_compute &&
+                       (c.src.space == 'M' || c.src.space == 'H' ||
+                        c.dst.space == 'M' || c.dst.space == 'H');
+      if (submits_kernel)
+        cmd_stream[i] = take(pool_kernel, (nk++) % (size_t)n_queues);
+      else if (host_side)
+        cmd_stream[i] = take(pool_hostcopy, (nh++) % (size_t)n_queues);
+      else
+        cmd_stream[i] = take(pool_devcopy, (nd++) % (size_t)n_queues);
+    }
   }
-  std::vector<hipStream_t> streams(pool.begin(), pool.begin() + need);
+  // unique streams actually in use (for the end-of-rep sync)
+  std::vector<hipStream_t> streams;
+  for (auto s : cmd_stream)
+    if (std::find(streams.begin(), streams.end(), s) == streams.end())
+      streams.push_back(s);
 
   std::vector<hipEvent_t> ev_start(ncmds), ev_stop(ncmds);
   if (enable_profiling) {
@@ -207,18 +239,13 @@ ConcResult conc_bench(const std::string& mode,
     for (int i = 0; i < ncmds; ++i)
       check_hip(hipEventCreateWithFlags(&join_ev[i], hipEventDisableTiming),
                 "join event");
-    hipStream_t master = streams[0];
+    hipStream_t master = master_stream;
     check_hip(hipStreamBeginCapture(master, hipStreamCaptureModeGlobal),
               "begin capture");
     check_hip(hipEventRecord(fork_ev, master), "record fork");
     for (int i = 0; i < ncmds; ++i) {
-      // stream[0] is the capture master; commands round-robin over the
-      // remaining streams so each gets its own graph branch when
-      // n_queues > ncmds (with one stream everything captures on master,
-      // which still yields independent nodes via the fork event).
-      hipStream_t s =
-          streams.size() > 1 ? streams[1 + (i % (streams.size() - 1))]
-                             : streams[0];
+      // each command captures on its typed stream -> its own graph branch
+      hipStream_t s = cmd_stream[i];
       check_hip(hipStreamWaitEvent(s, fork_ev, 0), "wait fork");
       cmds[i].submit(s);
       check_hip(hipEventRecord(join_ev[i], s), "record join");
@@ -233,10 +260,10 @@ ConcResult conc_bench(const std::string& mode,
   // graph upload / first-touch of managed memory should not pollute rep 0) ----
   {
     if (graph_mode) {
-      check_hip(hipGraphLaunch(graph_exec, streams[0]), "graph warmup");
-      check_hip(hipStreamSynchronize(streams[0]), "graph warmup sync");
+      check_hip(hipGraphLaunch(graph_exec, master_stream), "graph warmup");
+      check_hip(hipStreamSynchronize(master_stream), "graph warmup sync");
     } else {
-      for (int i = 0; i < ncmds; ++i) cmds[i].submit(streams[i % n_queues]);
+      for (int i = 0; i < ncmds; ++i) cmds[i].submit(cmd_stream[i]);
       check_hip(hipDeviceSynchronize(), "warmup sync");
     }
   }
@@ -248,26 +275,30 @@ ConcResult conc_bench(const std::string& mode,
     if (serial) {
       long total = 0;
       for (int i = 0; i < ncmds; ++i) {
+        // serial still uses the TYPED stream (sync-per-command keeps the
+        // semantics identical) so each command runs on its best engine —
+        // a serial H2D on a kernel-tainted stream would measure the blit
+        // path, not the SDMA path it gets in the concurrent run.
         long c0 = now_us();
-        if (enable_profiling) (void)hipEventRecord(ev_start[i], streams[0]);
-        cmds[i].submit(streams[0]);
-        if (enable_profiling) (void)hipEventRecord(ev_stop[i], streams[0]);
-        check_hip(hipStreamSynchronize(streams[0]), "serial sync");
+        if (enable_profiling) (void)hipEventRecord(ev_start[i], cmd_stream[i]);
+        cmds[i].submit(cmd_stream[i]);
+        if (enable_profiling) (void)hipEventRecord(ev_stop[i], cmd_stream[i]);
+        check_hip(hipStreamSynchronize(cmd_stream[i]), "serial sync");
         long c1 = now_us();
         res.per_cmd_us[i] = std::min(res.per_cmd_us[i], c1 - c0);
         total += c1 - c0;
       }
       min_total = std::min(min_total, total);
     } else if (graph_mode) {
-      check_hip(hipGraphLaunch(graph_exec, streams[0]), "graph launch");
-      check_hip(hipStreamSynchronize(streams[0]), "graph sync");
+      check_hip(hipGraphLaunch(graph_exec, master_stream), "graph launch");
+      check_hip(hipStreamSynchronize(master_stream), "graph sync");
       min_total = std::min(min_total, now_us() - t0);
     } else if (threads_mode) {
       std::vector<std::thread> ts;
       ts.reserve(ncmds);
       for (int i = 0; i < ncmds; ++i) {
         ts.emplace_back([&, i]() {
-          hipStream_t s = streams[i % n_queues];
+          hipStream_t s = cmd_stream[i];
           if (enable_profiling) (void)hipEventRecord(ev_start[i], s);
           cmds[i].submit(s);
           if (enable_profiling) (void)hipEventRecord(ev_stop[i], s);
@@ -278,13 +309,13 @@ ConcResult conc_bench(const std::string& mode,
       min_total = std::min(min_total, now_us() - t0);
     } else { // in_order / nowait
       for (int i = 0; i < ncmds; ++i) {
-        hipStream_t s = streams[i % n_queues];
+        hipStream_t s = cmd_stream[i];
         if (enable_profiling) (void)hipEventRecord(ev_start[i], s);
         cmds[i].submit(s);
         if (enable_profiling) (void)hipEventRecord(ev_stop[i], s);
       }
-      for (int q = 0; q < n_queues; ++q)
-        check_hip(hipStreamSynchronize(streams[q]), "stream sync");
+      for (auto s : streams)
+        check_hip(hipStreamSynchronize(s), "stream sync");
       min_total = std::min(min_total, now_us() - t0);
     }
 
