@@ -194,6 +194,17 @@ PYBIND11_MODULE(_hpk, m) {
                                         as_stream(stream)),
                          "hipMemcpyAsync");
         });
+  // kind: 1=H2D 2=D2H 3=D2D 4=default (hipMemcpyKind values) — engine
+  // selection differs between Default and explicit kinds on ROCm 7.2.
+  m.def("memcpy_async_kind",
+        [](uintptr_t dst, uintptr_t src, size_t nbytes, int kind,
+           uintptr_t stream) {
+          hpk::check_hip(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                                        reinterpret_cast<const void*>(src),
+                                        nbytes, (hipMemcpyKind)kind,
+                                        as_stream(stream)),
+                         "hipMemcpyAsync(kind)");
+        });
   m.def("device_synchronize", [] {
     py::gil_scoped_release release;
     hpk::check_hip(hipDeviceSynchronize(), "hipDeviceSynchronize");
