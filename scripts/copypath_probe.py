@@ -101,6 +101,40 @@ def main():
     print(f"{'duplex explicit fresh':24s} "
           f"{duplex(H2D, D2H, fresh_h2d, fresh_d2h):8.1f} GB/s")
 
+    # ---- allocation-history sequence (reproduces the conc-bench pattern:
+    # the first H buffer in a process copies at ~57 GB/s, later small H
+    # buffers allocated after big alloc/free cycles measured ~29 GB/s) ----
+    print("# allocation-history sequence (22.5 MB pinned buffers)")
+    small = 22_500_000
+
+    def rate_on(hbuf, dbuf, n, stream):
+        best = float("inf")
+        hpk.memcpy_async_kind(dbuf, hbuf, n, DEFAULT, stream.cuda_stream)
+        stream.synchronize()
+        for _ in range(5):
+            t0 = time.perf_counter()
+            hpk.memcpy_async_kind(dbuf, hbuf, n, DEFAULT, stream.cuda_stream)
+            stream.synchronize()
+            best = min(best, time.perf_counter() - t0)
+        return n / best / 1e9
+
+    s = torch.cuda.Stream()
+    h_a = hpk.host_malloc(small)
+    d_a = hpk.hip_malloc(small)
+    print(f"{'seq: first small H':24s} {rate_on(h_a, d_a, small, s):8.1f} GB/s")
+    # big alloc/free churn (what the earlier command lists did)
+    for _ in range(3):
+        hb = hpk.host_malloc(1 << 30)
+        db = hpk.hip_malloc(1 << 30)
+        hpk.memcpy_async_kind(db, hb, 1 << 30, DEFAULT, s.cuda_stream)
+        s.synchronize()
+        hpk.host_free(hb)
+        hpk.hip_free(db)
+    h_b = hpk.host_malloc(small)
+    d_b = hpk.hip_malloc(small)
+    print(f"{'seq: post-churn new H':24s} {rate_on(h_b, d_b, small, s):8.1f} GB/s")
+    print(f"{'seq: original H again':24s} {rate_on(h_a, d_a, small, s):8.1f} GB/s")
+
 
 if __name__ == "__main__":
     main()
