@@ -103,6 +103,18 @@ __global__ void k_copy_b16(const uint4* __restrict__ src, uint4* __restrict__ ds
   for (; i < n16; i += stride) dst[i] = src[i];
 }
 
+// Nontemporal variant: streaming load/store hints (slc) — tells the cache
+// hierarchy not to retain lines; candidate win for copies far beyond L3.
+__global__ void k_copy_b16_nt(const uint4* __restrict__ src,
+                              uint4* __restrict__ dst, size_t n16) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n16; i += stride) {
+    uint4 v = __builtin_nontemporal_load(src + i);
+    __builtin_nontemporal_store(v, dst + i);
+  }
+}
+
 // 4x-unrolled variant: 64 B per thread per iteration, 4 loads in flight
 // before the first store — deeper MLP for the HBM path at large sizes.
 __global__ void k_copy_b16x4(const uint4* __restrict__ src,
@@ -315,7 +327,10 @@ void launch_copy_kernel_tuned(void* dst, const void* src, size_t nbytes,
   size_t blocks = (n16 + kBlock - 1) / kBlock;
   if (blocks == 0) blocks = 1;
   if (blocks > grid_cap) blocks = grid_cap;
-  if (unroll >= 4) {
+  if (unroll == 5) { // nontemporal streaming variant
+    hipLaunchKernelGGL(k_copy_b16_nt, dim3(blocks), dim3(kBlock), 0, stream,
+                       (const uint4*)src, (uint4*)dst, n16);
+  } else if (unroll >= 4) {
     hipLaunchKernelGGL(k_copy_b16x4, dim3(blocks), dim3(kBlock), 0, stream,
                        (const uint4*)src, (uint4*)dst, n16);
   } else {

@@ -57,8 +57,8 @@ def main():
         variants = [("hipMemcpyAsync", lambda: hpk.memcpy_async(
             dst.data_ptr(), src.data_ptr(), nbytes,
             torch.cuda.current_stream().cuda_stream))]
-        for unroll in (1, 4):
-            for cap in (4096, 16384, 65536):
+        for unroll in (1, 4, 5):  # 5 = nontemporal streaming hints
+            for cap in (16384, 65536, 131072):
                 variants.append((
                     f"kernel u{unroll} cap{cap}",
                     lambda u=unroll, c=cap: hpk.copy_kernel_tuned(
