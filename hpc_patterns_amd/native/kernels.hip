@@ -105,13 +105,17 @@ __global__ void k_copy_b16(const uint4* __restrict__ src, uint4* __restrict__ ds
 
 // Nontemporal variant: streaming load/store hints (slc) — tells the cache
 // hierarchy not to retain lines; candidate win for copies far beyond L3.
+typedef __attribute__((ext_vector_type(4))) unsigned int uint4_ev_t;
+
 __global__ void k_copy_b16_nt(const uint4* __restrict__ src,
                               uint4* __restrict__ dst, size_t n16) {
+  const uint4_ev_t* s = (const uint4_ev_t*)src;
+  uint4_ev_t* d = (uint4_ev_t*)dst;
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < n16; i += stride) {
-    uint4 v = __builtin_nontemporal_load(src + i);
-    __builtin_nontemporal_store(v, dst + i);
+    uint4_ev_t v = __builtin_nontemporal_load(s + i);
+    __builtin_nontemporal_store(v, d + i);
   }
 }
 
