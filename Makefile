@@ -19,9 +19,9 @@ PYBIND_INC := $(shell $(PYTHON) -c "import pybind11; print(pybind11.get_include(
 EXT_SUFFIX := .so
 
 CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(GPU_ARCH) -I$(NATIVE) -Wall
-LDFLAGS    := -L/opt/rocm/lib -lrocm_smi64 -lrocprofiler-sdk-roctx
+LDFLAGS    := -L/opt/rocm/lib -lrocm_smi64 -lrocprofiler-sdk-roctx -lhsa-runtime64
 
-LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip $(NATIVE)/trace.hip
+LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip $(NATIVE)/trace.hip $(NATIVE)/sdma.hip
 LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
 
 EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)

@@ -64,6 +64,7 @@ class FlagshipPatternStep:
 
         self.ops = ops
         self.hpk = native()
+        self._copy_handles = []
         cfg = self.config
         dev = self.device
         torch.cuda.set_device(dev)
@@ -84,6 +85,10 @@ class FlagshipPatternStep:
                           torch.cuda.current_stream().cuda_stream)
         self.compute_out = torch.empty(cfg["compute_globalsize"],
                                        dtype=torch.float32, device=dev)
+        try:
+            self._sdma_engines = self.hpk.sdma_num_engines(dev.index or 0)
+        except Exception:
+            self._sdma_engines = 0
 
         self.distributed = (self.use_distributed and self.world_size > 1
                             and dist.is_initialized())
@@ -124,13 +129,35 @@ class FlagshipPatternStep:
         peer = self.rank + 1 if self.rank % 2 == 0 else self.rank - 1
         return peer if peer < self.world_size else None
 
-    def _h2d(self, stream) -> None:
-        self.hpk.memcpy_async(self.h2d_dev, self.h2d_host,
-                              self.config["h2d_bytes"], stream.cuda_stream)
+    # H2D/D2H go through EXPLICIT SDMA engines (engine 0 / engine 1): in a
+    # torch process rocclr routes pinned D2H hipMemcpyAsync through the
+    # shader-blit kernel, which steals CUs from the compute command and
+    # serializes against H2D (measured: profiles/copypath_r9). The explicit
+    # path returns a completion handle; _wait_copies() joins them.
+    def _h2d(self, stream=None) -> None:
+        if self._sdma_engines >= 1:
+            self._copy_handles.append(self.hpk.sdma_copy_begin(
+                self.h2d_dev, self.h2d_host, self.config["h2d_bytes"],
+                self.device.index or 0, 0))
+        else:
+            s = stream if stream is not None else torch.cuda.current_stream()
+            self.hpk.memcpy_async(self.h2d_dev, self.h2d_host,
+                                  self.config["h2d_bytes"], s.cuda_stream)
 
-    def _d2h(self, stream) -> None:
-        self.hpk.memcpy_async(self.d2h_host, self.d2h_dev,
-                              self.config["d2h_bytes"], stream.cuda_stream)
+    def _d2h(self, stream=None) -> None:
+        if self._sdma_engines >= 2:
+            self._copy_handles.append(self.hpk.sdma_copy_begin(
+                self.d2h_host, self.d2h_dev, self.config["d2h_bytes"],
+                self.device.index or 0, 1))
+        else:
+            s = stream if stream is not None else torch.cuda.current_stream()
+            self.hpk.memcpy_async(self.d2h_host, self.d2h_dev,
+                                  self.config["d2h_bytes"], s.cuda_stream)
+
+    def _wait_copies(self) -> None:
+        for h in self._copy_handles:
+            self.hpk.sdma_wait(h)
+        self._copy_handles.clear()
 
     # ---- calibration: make the compute command last about as long as the
     # slowest copy (linear model, reference autotuner main.cpp:226-258) ----
@@ -138,6 +165,7 @@ class FlagshipPatternStep:
         dev_sync = torch.cuda.synchronize
         cur = torch.cuda.current_stream()
         self._h2d(cur)  # first-touch warmup
+        self._wait_copies()
         dev_sync()
         t0 = time.perf_counter()
         self.ops.copy_kernel(self.d2d_dst, self.d2d_src)
@@ -145,6 +173,7 @@ class FlagshipPatternStep:
         t_d2d = time.perf_counter() - t0
         t0 = time.perf_counter()
         self._h2d(cur)
+        self._wait_copies()
         dev_sync()
         t_h2d = time.perf_counter() - t0
         target = max(t_d2d, t_h2d)
@@ -180,6 +209,7 @@ class FlagshipPatternStep:
                     r.wait()
             ar_work.wait()
 
+        self._wait_copies()
         for s in self.streams:
             s.synchronize()
 
@@ -212,6 +242,7 @@ class FlagshipPatternStep:
             ):
                 t0 = time.perf_counter()
                 fn()
+                self._wait_copies()
                 torch.cuda.synchronize()
                 times.append(time.perf_counter() - t0)
             return times
@@ -226,6 +257,7 @@ class FlagshipPatternStep:
                 self.ops.copy_kernel(self.d2d_dst, self.d2d_src, stream=s1)
             self._h2d(s2)
             self._d2h(s3)
+            self._wait_copies()
             for s in self.streams:
                 s.synchronize()
             return time.perf_counter() - t0

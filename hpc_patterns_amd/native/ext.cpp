@@ -155,6 +155,20 @@ PYBIND11_MODULE(_hpk, m) {
     hpk::ipc_close_handle(reinterpret_cast<void*>(dptr));
   });
   m.def("enable_peer_access", &hpk::enable_peer_access);
+  m.def("sdma_num_engines", &hpk::sdma_num_engines);
+  m.def("sdma_copy_begin",
+        [](uintptr_t dst, uintptr_t src, size_t nbytes, int device,
+           int engine_index) {
+          return reinterpret_cast<uintptr_t>(hpk::sdma_copy_begin(
+              reinterpret_cast<void*>(dst), reinterpret_cast<const void*>(src),
+              nbytes, device, engine_index));
+        },
+        py::arg("dst"), py::arg("src"), py::arg("nbytes"), py::arg("device") = 0,
+        py::arg("engine_index") = -1);
+  m.def("sdma_wait", [](uintptr_t handle) {
+    py::gil_scoped_release release;
+    hpk::sdma_wait(reinterpret_cast<void*>(handle));
+  });
   m.def("trace_push", [](const std::string& n) { hpk::trace_push(n.c_str()); });
   m.def("trace_pop", &hpk::trace_pop);
   m.def("trace_mark", [](const std::string& n) { hpk::trace_mark(n.c_str()); });

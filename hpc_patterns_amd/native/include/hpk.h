@@ -137,6 +137,19 @@ void* ipc_open_handle(const std::vector<uint8_t>& handle);
 void ipc_close_handle(void* dptr);
 
 // ---------------------------------------------------------------------------
+// Explicit SDMA-engine copies (sdma.hip) — hsa_amd_memory_async_copy[_on_
+// engine]: copies placed on a named DMA engine, bypassing rocclr's
+// blit-fallback heuristics (the reference's copy-engine selection knobs,
+// done natively).
+// ---------------------------------------------------------------------------
+int sdma_num_engines(int device);
+// engine_index < 0 lets ROCr pick. Returns a handle; copy completes when
+// sdma_wait(handle) returns (handle is consumed).
+void* sdma_copy_begin(void* dst, const void* src, size_t nbytes, int device,
+                      int engine_index);
+void sdma_wait(void* handle);
+
+// ---------------------------------------------------------------------------
 // Tracing (trace.hip) — roctx ranges for rocprofv3 --marker-trace.
 // ---------------------------------------------------------------------------
 void trace_push(const char* name);
