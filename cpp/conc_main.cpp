@@ -37,6 +37,7 @@ void help_and_exit(const std::string& bin, const std::string& msg) {
   std::cout
       << "Usage: " << bin << " " << hpk::allowed_modes << "\n"
       << "                [--enable_profiling] [--verbose] [--copy_kernel]\n"
+      << "                [--copy_engine auto|shader|sdma]\n"
       << "                [--tripcount_C <tripcount>]\n"
       << "                [--globalsize_{C,A2B} <global_size>]\n"
       << "                [--globalsize_default_memory <floats>]\n"
@@ -115,7 +116,8 @@ int main(int argc, char* argv[]) {
   std::map<std::string, long> params_cli = {{"globalsize_C", -1},
                                             {"tripcount_C", -1},
                                             {"globalsize_default_memory", -1}};
-  bool enable_profiling = false, verbose = false, use_copy_kernel = false;
+  bool enable_profiling = false, verbose = false;
+  int copy_engine = hpk::kCopyEngineAuto;
   int n_queues = -1, n_repetitions = 10;
   double min_bandwidth = -1;
   std::string csv_path;
@@ -133,7 +135,14 @@ int main(int argc, char* argv[]) {
     };
     if (s == "--enable_profiling") enable_profiling = true;
     else if (s == "--verbose") verbose = true;
-    else if (s == "--copy_kernel") use_copy_kernel = true;
+    else if (s == "--copy_kernel") copy_engine = hpk::kCopyEngineShader;
+    else if (s == "--copy_engine") {
+      std::string v = next("--copy_engine");
+      if (v == "auto") copy_engine = hpk::kCopyEngineAuto;
+      else if (v == "shader") copy_engine = hpk::kCopyEngineShader;
+      else if (v == "sdma") copy_engine = hpk::kCopyEngineSdma;
+      else help_and_exit(argv[0], "copy_engine must be auto|shader|sdma");
+    }
     else if (s == "--queues") n_queues = std::stoi(next("--queues"));
     else if (s == "--repetitions") n_repetitions = std::stoi(next("--repetitions"));
     else if (s == "--min_bandwidth") min_bandwidth = std::stod(next("--min_bandwidth"));
@@ -181,7 +190,7 @@ int main(int argc, char* argv[]) {
     std::cout << "# Performing Autotuning to Balance Commands Times" << std::endl;
     std::vector<std::string> uniq_vec(uniq.begin(), uniq.end());
     auto base = hpk::conc_bench("serial", uniq_vec, params, false, n_queues,
-                                n_repetitions, verbose, use_copy_kernel);
+                                n_repetitions, verbose, copy_engine);
     long target = std::numeric_limits<long>::max();
     for (size_t i = 0; i < uniq_vec.size(); ++i)
       if (uniq_vec[i] != "C")
@@ -223,7 +232,7 @@ int main(int argc, char* argv[]) {
 
     auto serial = hpk::conc_bench("serial", cmds, params, enable_profiling,
                                   n_queues, n_repetitions, verbose,
-                                  use_copy_kernel);
+                                  copy_engine);
     std::cout << "Minimum Measured Total Time Serial: " << serial.total_us
               << "us" << std::endl;
     for (size_t i = 0; i < cmds.size(); ++i) {
@@ -245,7 +254,7 @@ int main(int argc, char* argv[]) {
       std::cerr << "  WARNING: Large Unbalance Between Commands" << std::endl;
 
     auto conc = hpk::conc_bench(mode, cmds, params, enable_profiling, n_queues,
-                                n_repetitions, verbose, use_copy_kernel);
+                                n_repetitions, verbose, copy_engine);
     int bw_errno = 0;
     std::string conc_info =
         time_info(cmds, conc.total_us, params, min_bandwidth, &bw_errno);

@@ -33,7 +33,11 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--enable_profiling", action="store_true")
     p.add_argument("--verbose", action="store_true")
     p.add_argument("--copy_kernel", action="store_true",
-                   help="shader-blit copies instead of SDMA hipMemcpyAsync")
+                   help="legacy alias for --copy_engine shader")
+    p.add_argument("--copy_engine", choices=["auto", "shader", "sdma"],
+                   default="auto",
+                   help="copy path for A2B commands: runtime-picked | "
+                        "hand-written kernel | explicit SDMA engine")
     p.add_argument("--tripcount_C", type=int, default=-1)
     p.add_argument("--globalsize_C", type=int, default=-1)
     p.add_argument("--globalsize_default_memory", type=int, default=-1)
@@ -79,7 +83,8 @@ def main(argv=None) -> int:
     if any(auto_flags.values()) and len(all_cmds) > 1:
         print("# Performing Autotuning to Balance Commands Times")
         base = run_bench("serial", all_cmds, params, n_repetitions=args.repetitions,
-                         use_copy_kernel=args.copy_kernel)
+                         use_copy_kernel=args.copy_kernel,
+                         copy_engine=args.copy_engine)
         params = autotune_rescale(all_cmds, base["per_cmd_us"], params, auto_flags)
 
     print("Parameters used:")
@@ -95,7 +100,8 @@ def main(argv=None) -> int:
         serial = run_bench("serial", cmds, params,
                            enable_profiling=args.enable_profiling,
                            n_queues=args.queues, n_repetitions=args.repetitions,
-                           verbose=args.verbose, use_copy_kernel=args.copy_kernel)
+                           verbose=args.verbose, use_copy_kernel=args.copy_kernel,
+                           copy_engine=args.copy_engine)
         print(f"Minimum Measured Total Time Serial: {serial['total_us']}us")
         for i, c in enumerate(cmds):
             nbytes = params[f"globalsize_{c}"] * 4 if c != "C" else 0
@@ -109,7 +115,8 @@ def main(argv=None) -> int:
         conc = run_bench(args.mode, cmds, params,
                          enable_profiling=args.enable_profiling,
                          n_queues=args.queues, n_repetitions=args.repetitions,
-                         verbose=args.verbose, use_copy_kernel=args.copy_kernel)
+                         verbose=args.verbose, use_copy_kernel=args.copy_kernel,
+                         copy_engine=args.copy_engine)
         nbytes = sum(params[f"globalsize_{c}"] * 4 for c in cmds if c != "C")
         gbps = 1e-3 * nbytes / conc["total_us"] if nbytes else None
         print(f"Minimum Measured Total Time //: "

@@ -91,18 +91,29 @@ def autotune_rescale(unique_commands: Sequence[str],
     return new
 
 
+COPY_ENGINES = {"auto": 0, "shader": 1, "sdma": 2}
+
+
 def run_bench(mode: str, commands: Sequence[str],
               params: Mapping[str, int] | None = None,
               enable_profiling: bool = False, n_queues: int = -1,
               n_repetitions: int = 10, verbose: bool = False,
-              use_copy_kernel: bool = False) -> dict:
-    """Run the native engine. Requires the _hpk extension and a GPU."""
+              use_copy_kernel: bool = False,
+              copy_engine: str = "auto") -> dict:
+    """Run the native engine. Requires the _hpk extension and a GPU.
+
+    copy_engine: auto (hipMemcpyAsync) | shader (K2 copy kernel) | sdma
+    (explicit hsa_amd_memory_async_copy_on_engine). use_copy_kernel=True is
+    a legacy alias for copy_engine="shader"."""
     from .._native import native
 
     if mode not in ALLOWED_MODES:
         raise ValueError(f"mode '{mode}' not in {ALLOWED_MODES}")
+    if copy_engine not in COPY_ENGINES:
+        raise ValueError(f"copy_engine '{copy_engine}' not in {list(COPY_ENGINES)}")
+    engine = COPY_ENGINES["shader"] if use_copy_kernel else COPY_ENGINES[copy_engine]
     cmds = [validate_command(c) for c in commands]
     p = default_params(cmds, params)
     return native().conc_bench(mode, cmds, {k: int(v) for k, v in p.items()},
                                enable_profiling, n_queues, n_repetitions,
-                               verbose, use_copy_kernel)
+                               verbose, engine)

@@ -104,18 +104,38 @@ struct Command {
   size_t n_floats = 0;
   long tripcount = 0;
   long globalsize = 0;
-  bool shader_copy = false;
+  int copy_engine = kCopyEngineAuto;
+  int sdma_engine = -1; // explicit engine index for kCopyEngineSdma
+  mutable std::vector<void*> sdma_handles;
 
-  void submit(hipStream_t stream) const {
+  // SDMA path usable: both sides HIP-registered (not pageable M) and not
+  // inside a graph capture (HSA copies are not capturable).
+  bool sdma_ok() const {
+    return !is_compute && copy_engine == kCopyEngineSdma &&
+           src.space != 'M' && dst.space != 'M';
+  }
+
+  void submit(hipStream_t stream, bool in_graph = false) const {
     if (is_compute) {
       launch_busy_wait((float*)out.ptr, tripcount, globalsize, stream);
-    } else if (shader_copy && src.space != 'M' && dst.space != 'M') {
+    } else if (copy_engine == kCopyEngineShader && src.space != 'M' &&
+               dst.space != 'M') {
       launch_copy_kernel(dst.ptr, src.ptr, n_floats * sizeof(float), stream);
+    } else if (!in_graph && sdma_ok()) {
+      int dev = 0;
+      (void)hipGetDevice(&dev);
+      sdma_handles.push_back(sdma_copy_begin(
+          dst.ptr, src.ptr, n_floats * sizeof(float), dev, sdma_engine));
     } else {
       check_hip(hipMemcpyAsync(dst.ptr, src.ptr, n_floats * sizeof(float),
                                hipMemcpyDefault, stream),
                 "hipMemcpyAsync");
     }
+  }
+
+  void wait_sdma() const {
+    for (void* h : sdma_handles) sdma_wait(h);
+    sdma_handles.clear();
   }
 };
 
@@ -131,7 +151,7 @@ ConcResult conc_bench(const std::string& mode,
                       const std::vector<std::string>& commands,
                       const std::map<std::string, size_t>& params,
                       bool enable_profiling, int n_queues, int n_repetitions,
-                      bool verbose, bool use_copy_kernel) {
+                      bool verbose, int copy_engine) {
   if (!mode_is_allowed(mode))
     throw std::runtime_error("unknown mode '" + mode + "' (" + allowed_modes + ")");
 
@@ -148,7 +168,7 @@ ConcResult conc_bench(const std::string& mode,
   for (int i = 0; i < ncmds; ++i) {
     Command& c = cmds[i];
     c.name = commands[i];
-    c.shader_copy = use_copy_kernel;
+    c.copy_engine = copy_engine;
     if (c.name == "C") {
       c.is_compute = true;
       c.tripcount = (long)param(params, "tripcount_C", 40000);
@@ -193,11 +213,11 @@ ConcResult conc_bench(const std::string& mode,
     size_t nk = 0, nh = 0, nd = 0;
     for (int i = 0; i < ncmds; ++i) {
       const Command& c = cmds[i];
-      // a --copy_kernel shader copy SUBMITS a kernel -> kernel pool, so it
+      // a shader-engine copy SUBMITS a kernel -> kernel pool, so it
       // cannot taint a copy stream
       bool submits_kernel =
-          c.is_compute ||
-          (c.shader_copy && c.src.space != 'M' && c.dst.space != 'M');
+          c.is_compute || (c.copy_engine == kCopyEngineShader &&
+                           c.src.space != 'M' && c.dst.space != 'M');
       bool host_side = !c.is_compute &&
                        (c.src.space == 'M' || c.src.space == 'H' ||
                         c.dst.space == 'M' || c.dst.space == 'H');
@@ -208,6 +228,17 @@ ConcResult conc_bench(const std::string& mode,
       else
         cmd_stream[i] = take(pool_devcopy, (nd++) % (size_t)n_queues);
     }
+  }
+  // explicit SDMA engines: spread the copy commands round-robin over the
+  // device's engines so H2D and D2H land on different DMA queues
+  if (copy_engine == kCopyEngineSdma) {
+    int dev = 0;
+    (void)hipGetDevice(&dev);
+    int nengines = sdma_num_engines(dev);
+    int e = 0;
+    for (int i = 0; i < ncmds; ++i)
+      if (cmds[i].sdma_ok() && nengines > 0)
+        cmds[i].sdma_engine = (e++) % nengines;
   }
   // unique streams actually in use (for the end-of-rep sync)
   std::vector<hipStream_t> streams;
@@ -247,7 +278,7 @@ ConcResult conc_bench(const std::string& mode,
       // each command captures on its typed stream -> its own graph branch
       hipStream_t s = cmd_stream[i];
       check_hip(hipStreamWaitEvent(s, fork_ev, 0), "wait fork");
-      cmds[i].submit(s);
+      cmds[i].submit(s, /*in_graph=*/true);
       check_hip(hipEventRecord(join_ev[i], s), "record join");
       check_hip(hipStreamWaitEvent(master, join_ev[i], 0), "wait join");
     }
@@ -264,6 +295,7 @@ ConcResult conc_bench(const std::string& mode,
       check_hip(hipStreamSynchronize(master_stream), "graph warmup sync");
     } else {
       for (int i = 0; i < ncmds; ++i) cmds[i].submit(cmd_stream[i]);
+      for (int i = 0; i < ncmds; ++i) cmds[i].wait_sdma();
       check_hip(hipDeviceSynchronize(), "warmup sync");
     }
   }
@@ -283,6 +315,7 @@ ConcResult conc_bench(const std::string& mode,
         if (enable_profiling) (void)hipEventRecord(ev_start[i], cmd_stream[i]);
         cmds[i].submit(cmd_stream[i]);
         if (enable_profiling) (void)hipEventRecord(ev_stop[i], cmd_stream[i]);
+        cmds[i].wait_sdma();
         check_hip(hipStreamSynchronize(cmd_stream[i]), "serial sync");
         long c1 = now_us();
         res.per_cmd_us[i] = std::min(res.per_cmd_us[i], c1 - c0);
@@ -302,6 +335,7 @@ ConcResult conc_bench(const std::string& mode,
           if (enable_profiling) (void)hipEventRecord(ev_start[i], s);
           cmds[i].submit(s);
           if (enable_profiling) (void)hipEventRecord(ev_stop[i], s);
+          cmds[i].wait_sdma();
           check_hip(hipStreamSynchronize(s), "thread sync");
         });
       }
@@ -314,6 +348,7 @@ ConcResult conc_bench(const std::string& mode,
         cmds[i].submit(s);
         if (enable_profiling) (void)hipEventRecord(ev_stop[i], s);
       }
+      for (int i = 0; i < ncmds; ++i) cmds[i].wait_sdma();
       for (auto s : streams)
         check_hip(hipStreamSynchronize(s), "stream sync");
       min_total = std::min(min_total, now_us() - t0);

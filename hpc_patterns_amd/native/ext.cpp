@@ -21,12 +21,12 @@ py::dict conc_bench_py(const std::string& mode,
                        const std::vector<std::string>& commands,
                        const std::map<std::string, size_t>& params,
                        bool enable_profiling, int n_queues, int n_repetitions,
-                       bool verbose, bool use_copy_kernel) {
+                       bool verbose, int copy_engine) {
   hpk::ConcResult r;
   {
     py::gil_scoped_release release;
     r = hpk::conc_bench(mode, commands, params, enable_profiling, n_queues,
-                        n_repetitions, verbose, use_copy_kernel);
+                        n_repetitions, verbose, copy_engine);
   }
   py::dict d;
   d["total_us"] = r.total_us;
@@ -106,7 +106,7 @@ PYBIND11_MODULE(_hpk, m) {
   m.def("conc_bench", &conc_bench_py, py::arg("mode"), py::arg("commands"),
         py::arg("params"), py::arg("enable_profiling") = false,
         py::arg("n_queues") = -1, py::arg("n_repetitions") = 10,
-        py::arg("verbose") = false, py::arg("use_copy_kernel") = false);
+        py::arg("verbose") = false, py::arg("copy_engine") = 0);
 
   // ---- topology ----
   m.def("device_count", [] {

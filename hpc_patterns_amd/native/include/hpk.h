@@ -88,6 +88,14 @@ long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream);
 extern const std::string allowed_modes;
 bool mode_is_allowed(const std::string& mode);
 
+// copy-engine selection for A2B commands (the reference's copy-engine env
+// knobs as a first-class CLI switch):
+//   auto   -> hipMemcpyAsync (runtime picks SDMA or blit)
+//   shader -> hand-written K2 copy kernel (CU path)
+//   sdma   -> explicit hsa_amd_memory_async_copy_on_engine (DMA path)
+enum CopyEngine { kCopyEngineAuto = 0, kCopyEngineShader = 1,
+                  kCopyEngineSdma = 2 };
+
 struct ConcResult {
   long total_us = 0;                  // min over repetitions
   std::vector<long> per_cmd_us;       // serial mode: min per-command wall time
@@ -102,7 +110,7 @@ ConcResult conc_bench(const std::string& mode,
                       const std::vector<std::string>& commands,
                       const std::map<std::string, size_t>& params,
                       bool enable_profiling, int n_queues, int n_repetitions,
-                      bool verbose, bool use_copy_kernel);
+                      bool verbose, int copy_engine);
 
 // ---------------------------------------------------------------------------
 // Topology (topo.hip) — xGMI link discovery (reference p2p/topology.cpp).

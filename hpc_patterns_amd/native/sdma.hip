@@ -89,14 +89,36 @@ void check_hsa(hsa_status_t s, const char* what) {
 
 } // namespace
 
+namespace {
+// Engine mask for copies src_agent -> dst_agent. The status API wants the
+// REAL agent pair (gpu,gpu returns nothing useful on MI355X).
+uint32_t engine_mask_for(hsa_agent_t dst_agent, hsa_agent_t src_agent) {
+  uint32_t mask = 0;
+  if (hsa_amd_memory_copy_engine_status(dst_agent, src_agent, &mask) !=
+      HSA_STATUS_SUCCESS)
+    return 0;
+  return mask;
+}
+
+// engine_index-th set bit of mask (wrapping); 0 if mask empty.
+uint32_t pick_engine_bit(uint32_t mask, int engine_index) {
+  int n = __builtin_popcount(mask);
+  if (n == 0) return 0;
+  int want = engine_index % n;
+  uint32_t m = mask;
+  for (int i = 0; i < want; ++i) m &= m - 1; // drop lowest set bits
+  return m & ~(m - 1);                       // lowest remaining set bit
+}
+} // namespace
+
 int sdma_num_engines(int device) {
   auto& a = agents();
-  if (device < 0 || device >= (int)a.gpus.size()) return 0;
-  uint32_t mask = 0;
-  hsa_status_t s = hsa_amd_memory_copy_engine_status(a.gpus[device],
-                                                     a.gpus[device], &mask);
-  if (s != HSA_STATUS_SUCCESS) return 0;
-  return __builtin_popcount(mask);
+  if (device < 0 || device >= (int)a.gpus.size() || !a.have_cpu) return 0;
+  // H2D engines (dst=gpu, src=cpu) and D2H engines (dst=cpu, src=gpu)
+  uint32_t h2d = engine_mask_for(a.gpus[device], a.cpu);
+  uint32_t d2h = engine_mask_for(a.cpu, a.gpus[device]);
+  int nh = __builtin_popcount(h2d), nd = __builtin_popcount(d2h);
+  return nh > nd ? nh : nd;
 }
 
 // Begin an explicit-engine async copy; returns an opaque handle to wait on.
@@ -115,13 +137,17 @@ void* sdma_copy_begin(void* dst, const void* src, size_t nbytes, int device,
   auto* signal = new hsa_signal_t;
   check_hsa(hsa_signal_create(1, 0, nullptr, signal), "hsa_signal_create");
 
-  hsa_status_t s;
+  hsa_status_t s = HSA_STATUS_ERROR;
   if (engine_index >= 0) {
-    s = hsa_amd_memory_async_copy_on_engine(
-        dst, dst_agent, src, src_agent, nbytes, 0, nullptr, *signal,
-        (hsa_amd_sdma_engine_id_t)(1u << engine_index),
-        /*force_copy_on_sdma=*/true);
-  } else {
+    uint32_t bit =
+        pick_engine_bit(engine_mask_for(dst_agent, src_agent), engine_index);
+    if (bit != 0) {
+      s = hsa_amd_memory_async_copy_on_engine(
+          dst, dst_agent, src, src_agent, nbytes, 0, nullptr, *signal,
+          (hsa_amd_sdma_engine_id_t)bit, /*force_copy_on_sdma=*/true);
+    }
+  }
+  if (s != HSA_STATUS_SUCCESS) { // no engine requested/available: plain copy
     s = hsa_amd_memory_async_copy(dst, dst_agent, src, src_agent, nbytes, 0,
                                   nullptr, *signal);
   }
