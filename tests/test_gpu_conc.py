@@ -91,3 +91,10 @@ def test_copy_kernel_engine(run_bench):
     res = run_bench("in_order", ["D2D"], SMALL, use_copy_kernel=True,
                     n_repetitions=2)
     assert res["total_us"] > 0
+
+
+@pytest.mark.parametrize("engine", ["auto", "shader", "sdma"])
+def test_copy_engines(run_bench, engine):
+    res = run_bench("in_order", ["H2D", "D2H"], SMALL, n_repetitions=3,
+                    copy_engine=engine)
+    assert res["total_us"] > 0
