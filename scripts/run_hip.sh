@@ -27,10 +27,12 @@ do
     (
     export $envs
     echo "export $envs"
+    for engine in auto sdma; do
     for mode in in_order graph host_threads; do
         args=""
         for c in "${LCOMMANDS[@]}"; do args+=" --commands $c"; done
-        $BIN "$mode" --repetitions 5 $args
+        $BIN "$mode" --copy_engine "$engine" --repetitions 5 $args
+    done
     done
     ) 2>&1 | tee -a "$LOG"
 done
