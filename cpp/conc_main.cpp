@@ -37,7 +37,7 @@ void help_and_exit(const std::string& bin, const std::string& msg) {
   std::cout
       << "Usage: " << bin << " " << hpk::allowed_modes << "\n"
       << "                [--enable_profiling] [--verbose] [--copy_kernel]\n"
-      << "                [--copy_engine auto|shader|sdma]\n"
+      << "                [--copy_engine auto|shader|sdma] [--mfma]\n"
       << "                [--tripcount_C <tripcount>]\n"
       << "                [--globalsize_{C,A2B} <global_size>]\n"
       << "                [--globalsize_default_memory <floats>]\n"
@@ -134,6 +134,7 @@ int main(int argc, char* argv[]) {
       return args[i];
     };
     if (s == "--enable_profiling") enable_profiling = true;
+    else if (s == "--mfma") params_cli["payload_C_mfma"] = 1;
     else if (s == "--verbose") verbose = true;
     else if (s == "--copy_kernel") copy_engine = hpk::kCopyEngineShader;
     else if (s == "--copy_engine") {

@@ -114,6 +114,10 @@ def run_bench(mode: str, commands: Sequence[str],
     engine = COPY_ENGINES["shader"] if use_copy_kernel else COPY_ENGINES[copy_engine]
     cmds = [validate_command(c) for c in commands]
     p = default_params(cmds, params)
+    # pass through engine-level extras (e.g. payload_C_mfma) untouched
+    for k, v in dict(params or {}).items():
+        if k not in p and v != -1:
+            p[k] = int(v)
     return native().conc_bench(mode, cmds, {k: int(v) for k, v in p.items()},
                                enable_profiling, n_queues, n_repetitions,
                                verbose, engine)

@@ -104,6 +104,7 @@ struct Command {
   size_t n_floats = 0;
   long tripcount = 0;
   long globalsize = 0;
+  bool mfma_payload = false;
   int copy_engine = kCopyEngineAuto;
   int sdma_engine = -1; // explicit engine index for kCopyEngineSdma
   mutable std::vector<void*> sdma_handles;
@@ -117,7 +118,13 @@ struct Command {
 
   void submit(hipStream_t stream, bool in_graph = false) const {
     if (is_compute) {
-      launch_busy_wait((float*)out.ptr, tripcount, globalsize, stream);
+      if (mfma_payload) {
+        // matrix-core busy payload: one wave per 64 work-items
+        launch_busy_wait_mfma((float*)out.ptr, tripcount,
+                              std::max<long>(globalsize / 64, 1), stream);
+      } else {
+        launch_busy_wait((float*)out.ptr, tripcount, globalsize, stream);
+      }
     } else if (copy_engine == kCopyEngineShader && src.space != 'M' &&
                dst.space != 'M') {
       launch_copy_kernel(dst.ptr, src.ptr, n_floats * sizeof(float), stream);
@@ -173,7 +180,12 @@ ConcResult conc_bench(const std::string& mode,
       c.is_compute = true;
       c.tripcount = (long)param(params, "tripcount_C", 40000);
       c.globalsize = (long)param(params, "globalsize_C", 1);
-      c.out.alloc('D', std::max<long>(c.globalsize, 1) * sizeof(float));
+      c.mfma_payload = param(params, "payload_C_mfma", 0) != 0;
+      // mfma kernel writes ceil(waves*64/256)*256 floats
+      size_t out_floats = c.mfma_payload
+          ? ((std::max<long>(c.globalsize / 64, 1) * 64 + 255) / 256) * 256
+          : std::max<long>(c.globalsize, 1);
+      c.out.alloc('D', out_floats * sizeof(float));
     } else {
       if (c.name.size() != 2)
         throw std::runtime_error("bad command '" + c.name + "'");
