@@ -164,9 +164,6 @@ def test_dist_pattern(world, fn, dist_env):
 
 
 def body_sweep(rank, world):
-    import io
-    from contextlib import redirect_stdout
-
     from hpc_patterns_amd.parallel import sweep
 
     t = sweep.bench_algo("ring", 4096, iters=2,
