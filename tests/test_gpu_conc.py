@@ -61,11 +61,16 @@ def test_overlap_compute_copy(run_bench):
 
 def test_graph_mode_concurrent(run_bench):
     # compute || copy on independent graph branches must overlap (two D2D
-    # copies would NOT — both are HBM-bandwidth-bound, no speedup to find)
-    base = run_bench("serial", ["C", "D2D"], SMALL, n_repetitions=3)
+    # copies would NOT — both are HBM-bandwidth-bound, no speedup to find).
+    # ms-scale commands: at the SMALL 64 MB scale the ~20 µs graph-launch
+    # overhead swamps the overlap (observed flaky).
+    big = dict(SMALL)
+    big["globalsize_DD"] = 1 << 27  # 512 MB -> ~200 µs copy
+    big["globalsize_C"] = 1 << 18
+    base = run_bench("serial", ["C", "D2D"], big, n_repetitions=3)
     t_c, t_copy = base["per_cmd_us"]
-    params = dict(SMALL)
-    params["tripcount_C"] = max(int(SMALL["tripcount_C"] * t_copy / max(t_c, 1)), 1)
+    params = dict(big)
+    params["tripcount_C"] = max(int(big["tripcount_C"] * t_copy / max(t_c, 1)), 1)
     serial = run_bench("serial", ["C", "D2D"], params, n_repetitions=5)
     graph = run_bench("graph", ["C", "D2D"], params, n_repetitions=5)
     speedup = serial["total_us"] / max(graph["total_us"], 1)
