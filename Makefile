@@ -67,8 +67,9 @@ $(BIN):
 # Race/memory-error hunting build: host-side ASan on the
 # binaries. Run on a GPU box with ASAN_OPTIONS=detect_leaks=0 (the HIP
 # runtime intentionally holds allocations).
-asan: CXXFLAGS += -fsanitize=address -g1
-asan: clean-bins $(BINARIES)
+asan:
+	$(MAKE) clean-bins
+	$(MAKE) CXXFLAGS="$(CXXFLAGS) -fsanitize=address -g1" bins
 
 clean-bins:
 	rm -rf $(BIN) $(filter %_main.o,$(wildcard $(BUILD)/*.o))

@@ -26,8 +26,8 @@ for envs in "HIP_VISIBLE_DEVICES=0" \
 do
     (
     export $envs
-    echo "export $envs"
     for engine in auto sdma; do
+    echo "export $envs HPK_COPY_ENGINE=$engine"
     for mode in in_order graph host_threads; do
         args=""
         for c in "${LCOMMANDS[@]}"; do args+=" --commands $c"; done
