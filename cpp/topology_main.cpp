@@ -74,6 +74,15 @@ int main(int argc, char* argv[]) {
         std::printf("gpu%d <-> gpu%d : %ld - %ld MB/s (weight %ld)\n", i, j,
                     m[i][j].min_bw_mbps, m[i][j].max_bw_mbps, m[i][j].weight);
 
+  auto parts = hpk::partition_info();
+  if (!parts.empty()) {
+    std::printf("# partition modes (compute/memory):\n");
+    for (size_t i = 0; i < parts.size(); ++i)
+      std::printf("gpu%zu: %s / %s\n", i,
+                  parts[i].compute.empty() ? "?" : parts[i].compute.c_str(),
+                  parts[i].memory.empty() ? "?" : parts[i].memory.c_str());
+  }
+
   std::printf("# connectivity planes (direct-P2P reachability):\n");
   for (size_t p = 0; p < planes.size(); ++p) {
     std::printf("plane %zu:", p);

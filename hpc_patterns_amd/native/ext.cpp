@@ -140,6 +140,21 @@ PYBIND11_MODULE(_hpk, m) {
     py::gil_scoped_release release;
     return hpk::p2p_planes();
   });
+  m.def("partition_info", [] {
+    std::vector<hpk::PartitionInfo> v;
+    {
+      py::gil_scoped_release release;
+      v = hpk::partition_info();
+    }
+    py::list out;
+    for (auto& pi : v) {
+      py::dict d;
+      d["compute"] = pi.compute;
+      d["memory"] = pi.memory;
+      out.append(d);
+    }
+    return out;
+  });
 
   // ---- IPC / peer transport ----
   m.def("ipc_get_handle", [](uintptr_t dptr) {

@@ -128,6 +128,14 @@ struct LinkInfo {
 int device_count();
 // NxN matrix; [i][i] is zero-initialized LinkInfo.
 std::vector<std::vector<LinkInfo>> link_matrix();
+
+// XCD compute-partition (SPX/DPX/CPX...) and memory-partition (NPS*) mode
+// per device — the MI355X analog of the reference's tile-fission awareness.
+struct PartitionInfo {
+  std::string compute; // e.g. "SPX", "CPX"; empty if unavailable
+  std::string memory;  // e.g. "NPS1"; empty if unavailable
+};
+std::vector<PartitionInfo> partition_info();
 // Connected components under direct-P2P reachability (the reference's
 // "connectivity planes", topology.cpp:76-89). MI355X nodes are fully
 // connected, so this is usually one plane — the interesting data is the

@@ -75,6 +75,32 @@ std::vector<std::vector<LinkInfo>> link_matrix() {
   return m;
 }
 
+std::vector<PartitionInfo> partition_info() {
+  // XCD/memory partition modes (SPX/DPX/..., NPS1/NPS4) — the MI355X
+  // replacement for the reference's tile-fission awareness
+  // (devices.hpp:30-34): a CPX-partitioned GPU shows up as several HIP
+  // devices, so placement policies need to know.
+  std::vector<PartitionInfo> out;
+  if (rsmi_init(0) != RSMI_STATUS_SUCCESS) return out;
+  uint32_t n = 0;
+  if (rsmi_num_monitor_devices(&n) == RSMI_STATUS_SUCCESS) {
+    for (uint32_t i = 0; i < n; ++i) {
+      PartitionInfo pi;
+      char buf[64] = {0};
+      if (rsmi_dev_compute_partition_get(i, buf, sizeof(buf)) ==
+          RSMI_STATUS_SUCCESS)
+        pi.compute = buf;
+      char mbuf[64] = {0};
+      if (rsmi_dev_memory_partition_get(i, mbuf, sizeof(mbuf)) ==
+          RSMI_STATUS_SUCCESS)
+        pi.memory = mbuf;
+      out.push_back(pi);
+    }
+  }
+  rsmi_shut_down();
+  return out;
+}
+
 std::vector<std::vector<int>> p2p_planes() {
   int n = device_count();
   auto m = link_matrix();
