@@ -184,3 +184,21 @@ def test_sdma_explicit_engine_copy(ops, dev):
     for p in (host, host2):
         hpk.host_free(p)
     hpk.hip_free(dev_buf)
+
+
+def test_fill_large_nontemporal_path(ops, dev):
+    # >32 MB exercises the nontemporal-store fill variant
+    n = (64 << 20) // 4 + 3
+    t = torch.empty(n, dtype=torch.float32, device=dev)
+    ops.fill(t, -2.25)
+    torch.cuda.synchronize()
+    assert torch.all(t == -2.25)
+
+
+def test_copy_large_nontemporal_path(ops, dev):
+    n = (64 << 20) // 4 + 5
+    src = torch.rand(n, device=dev)
+    dst = torch.zeros_like(src)
+    ops.copy_kernel(dst, src)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, src)
