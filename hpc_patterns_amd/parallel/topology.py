@@ -82,3 +82,32 @@ def topology_order(n_gpus: int) -> list[int]:
     except Exception:
         pass
     return list(range(n_gpus))
+
+
+def _main() -> int:  # python -m hpc_patterns_amd.parallel.topology
+    try:
+        m = link_matrix()
+    except Exception as e:
+        print(f"no GPU / native core: {e}")
+        return 1
+    n = len(m)
+    print(f"# {n} HIP device(s)")
+    for i in range(n):
+        row = " ".join(
+            "-" if i == j else
+            f"p2p={m[i][j]['p2p']},t={m[i][j]['link_type']},w={m[i][j]['weight']}"
+            for j in range(n))
+        print(f"gpu{i}: {row}")
+    print("planes:", planes_from_matrix(m))
+    print("order:", topology_order_from_matrix(m))
+    try:
+        from .._native import native
+
+        print("partitions:", native().partition_info())
+    except Exception:
+        pass
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(_main())
