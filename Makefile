@@ -19,7 +19,8 @@ PYBIND_INC := $(shell $(PYTHON) -c "import pybind11; print(pybind11.get_include(
 EXT_SUFFIX := .so
 
 CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(GPU_ARCH) -I$(NATIVE) -Wall
-LDFLAGS    := -L/opt/rocm/lib -lrocm_smi64 -lrocprofiler-sdk-roctx -lhsa-runtime64
+LDEXTRA    ?=
+LDFLAGS    := $(LDEXTRA) -L/opt/rocm/lib -lrocm_smi64 -lrocprofiler-sdk-roctx -lhsa-runtime64
 
 LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip $(NATIVE)/trace.hip $(NATIVE)/sdma.hip
 LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
@@ -69,7 +70,7 @@ $(BIN):
 # runtime intentionally holds allocations).
 asan:
 	$(MAKE) clean-bins
-	$(MAKE) CXXFLAGS="$(CXXFLAGS) -fsanitize=address -g1" bins
+	$(MAKE) CXXFLAGS="$(CXXFLAGS) -fsanitize=address -g1" LDEXTRA="-fsanitize=address" bins
 
 clean-bins:
 	rm -rf $(BIN) $(filter %_main.o,$(wildcard $(BUILD)/*.o))
