@@ -158,6 +158,10 @@ def body_flagship_cpu_accounting(rank, world):
     (2, "body_pairwise_bandwidth"),
     (2, "body_pingpong"),
     (2, "body_flagship_cpu_accounting"),
+    # the driver's 8-rank node shape, rehearsed on CPU
+    (8, "body_ring"),
+    (8, "body_ring_rsag"),
+    (8, "body_pairwise"),
 ])
 def test_dist_pattern(world, fn, dist_env):
     run_dist(world, fn, int(dist_env["MASTER_PORT"]))
