@@ -171,6 +171,7 @@ PYBIND11_MODULE(_hpk, m) {
   });
   m.def("enable_peer_access", &hpk::enable_peer_access);
   m.def("sdma_num_engines", &hpk::sdma_num_engines);
+  m.def("sdma_num_engines_pair", &hpk::sdma_num_engines_pair);
   m.def("sdma_copy_begin",
         [](uintptr_t dst, uintptr_t src, size_t nbytes, int device,
            int engine_index) {

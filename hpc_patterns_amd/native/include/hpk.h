@@ -159,6 +159,8 @@ void ipc_close_handle(void* dptr);
 // done natively).
 // ---------------------------------------------------------------------------
 int sdma_num_engines(int device);
+// engines usable for src_device -> dst_device peer copies (xGMI SDMA).
+int sdma_num_engines_pair(int dst_device, int src_device);
 // engine_index < 0 lets ROCr pick. Returns a handle; copy completes when
 // sdma_wait(handle) returns (handle is consumed).
 void* sdma_copy_begin(void* dst, const void* src, size_t nbytes, int device,

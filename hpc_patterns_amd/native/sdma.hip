@@ -111,6 +111,17 @@ uint32_t pick_engine_bit(uint32_t mask, int engine_index) {
 }
 } // namespace
 
+int sdma_num_engines_pair(int dst_device, int src_device) {
+  // engines usable for src_device -> dst_device copies (xGMI SDMA engines
+  // for peer pairs: MI355X exposes 14 per GPU besides the 2 host engines)
+  auto& a = agents();
+  if (dst_device < 0 || dst_device >= (int)a.gpus.size() || src_device < 0 ||
+      src_device >= (int)a.gpus.size())
+    return 0;
+  return __builtin_popcount(
+      engine_mask_for(a.gpus[dst_device], a.gpus[src_device]));
+}
+
 int sdma_num_engines(int device) {
   auto& a = agents();
   if (device < 0 || device >= (int)a.gpus.size() || !a.have_cpu) return 0;
