@@ -202,3 +202,12 @@ def test_copy_large_nontemporal_path(ops, dev):
     ops.copy_kernel(dst, src)
     torch.cuda.synchronize()
     assert torch.equal(dst, src)
+
+
+def test_sdma_engine_pair_query(ops, dev):
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    # same-device pair: must not crash; >=0 engines
+    n = hpk.sdma_num_engines_pair(0, 0)
+    assert n >= 0
