@@ -234,6 +234,12 @@ def main() -> int:
     components = {}
     if not args.cpu:
         overlap = step.measure_overlap(reps=3)
+        if world > 1:
+            # report the WORST rank's overlap efficiency, not rank 0's
+            t = torch.tensor([overlap["overlap_efficiency"]],
+                             dtype=torch.float64, device=device)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            overlap["overlap_efficiency"] = float(t.item())
         components["stream_overlap_pct"] = round(
             100.0 * overlap["overlap_efficiency"], 1)
         components["stream_overlap_speedup"] = round(overlap["speedup"], 3)
