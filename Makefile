@@ -26,7 +26,7 @@ LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATI
 LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
 
 EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)
-BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p $(BIN)/hpk_interop
+BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p $(BIN)/hpk_interop $(BIN)/hpk_membench
 
 .PHONY: all ext bins clean
 all: ext bins
@@ -59,6 +59,9 @@ $(BIN)/hpk_p2p: $(BUILD)/p2p_main.o $(LIB_OBJS) | $(BIN)
 
 $(BIN)/hpk_interop: $(BUILD)/interop_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) -lrccl
+
+$(BIN)/hpk_membench: $(BUILD)/membench_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS)
 
 $(BUILD):
 	mkdir -p $(BUILD)

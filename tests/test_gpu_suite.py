@@ -213,3 +213,22 @@ def test_bin_allreduce_int_dtype():
                 "-n", "1", "--algo", "rccl", "-t", "int"])
     assert res.returncode == 0, res.stdout + res.stderr
     assert "Passed rank 0" in res.stdout and "dtype=int" in res.stdout
+
+
+def test_bin_membench_quick():
+    res = _run([str(REPO / "bin/hpk_membench"), "--quick"])
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "TFLOP/s bf16" in res.stdout and "GB/s payload" in res.stdout
+
+
+def test_bin_conc_min_bandwidth_floor():
+    # floor respected end to end: an impossible floor must FAIL (exit 1),
+    # a trivial floor must not trip the bandwidth check
+    res = _run([str(REPO / "bin/hpk_conc"), "in_order", "--repetitions", "3",
+                "--globalsize_default_memory", str(1 << 24),
+                "--min_bandwidth", "999999", "--commands", "D2D", "D2D"])
+    assert res.returncode == 1 and "Minimum Bandwidth not reached" in res.stdout
+    res2 = _run([str(REPO / "bin/hpk_conc"), "in_order", "--repetitions", "3",
+                 "--globalsize_default_memory", str(1 << 24),
+                 "--min_bandwidth", "0.001", "--commands", "C", "D2D"])
+    assert "Minimum Bandwidth not reached" not in res2.stdout
