@@ -92,6 +92,13 @@ PYBIND11_MODULE(_hpk, m) {
                               as_stream(stream));
         },
         py::arg("dst"), py::arg("src"), py::arg("n"), py::arg("stream") = 0);
+  m.def("acc_f32_nt",
+        [](uintptr_t dst, uintptr_t src, size_t n, uintptr_t stream) {
+          hpk::launch_acc_f32_nt(reinterpret_cast<float*>(dst),
+                                 reinterpret_cast<const float*>(src), n,
+                                 as_stream(stream));
+        },
+        py::arg("dst"), py::arg("src"), py::arg("n"), py::arg("stream") = 0);
   m.def("reduce_sum_f32",
         [](uintptr_t src, size_t n, uintptr_t stream) {
           py::gil_scoped_release release;

@@ -60,6 +60,9 @@ void launch_iota_f32(float* dst, size_t n, hipStream_t stream);
 // K3 accumulate: dst[i] += src[i] (reference Accumulate,
 // allreduce-mpi-sycl.cpp:27-31), vectorized float4 grid-stride.
 void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream);
+// nontemporal variant (read-once src / rmw-once dst, beyond-L3 sizes)
+void launch_acc_f32_nt(float* dst, const float* src, size_t n,
+                       hipStream_t stream);
 
 // Exact double-precision sum of n floats. Synchronizes `stream`.
 // Replaces the reference's O(N log N) host sort+sum checksum

@@ -203,6 +203,21 @@ __global__ void k_acc_f1(float* __restrict__ dst, const float* __restrict__ src,
   for (; i < n; i += stride) dst[i] += src[i];
 }
 
+// NT variant: src is read once (never re-read) -> nontemporal load; dst is
+// read-modify-write with an NT store (the line won't be revisited either).
+__global__ void k_acc_f4_nt(float4* __restrict__ dst,
+                            const float4* __restrict__ src, size_t n4) {
+  const f4_ev_t* s = (const f4_ev_t*)src;
+  f4_ev_t* d = (f4_ev_t*)dst;
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    f4_ev_t a = __builtin_nontemporal_load(d + i);
+    f4_ev_t b = __builtin_nontemporal_load(s + i);
+    __builtin_nontemporal_store(a + b, d + i);
+  }
+}
+
 // --------------------------------------------------------------------------
 // int32 twins of fill/accumulate (the reference instantiates its miniapps
 // for float AND int via -DAPP_DATA_TYPE, mpi-sycl/CMakeLists.txt:4-5).
@@ -492,6 +507,24 @@ long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream) {
   long long s = 0;
   for (long long v : h) s += v;
   return s;
+}
+
+void launch_acc_f32_nt(float* dst, const float* src, size_t n,
+                       hipStream_t stream) {
+  if (((uintptr_t)dst % 16 == 0) && ((uintptr_t)src % 16 == 0) && n >= 4) {
+    size_t n4 = n / 4;
+    size_t tail = n - n4 * 4;
+    size_t blocks = (n4 + kBlock - 1) / kBlock;
+    if (blocks > 131072) blocks = 131072;
+    hipLaunchKernelGGL(k_acc_f4_nt, dim3(blocks), dim3(kBlock), 0, stream,
+                       (float4*)dst, (const float4*)src, n4);
+    if (tail)
+      hipLaunchKernelGGL(k_acc_f1, dim3(1), dim3(64), 0, stream, dst + n4 * 4,
+                         src + n4 * 4, tail);
+    check_hip(hipGetLastError(), "launch_acc_f32_nt");
+  } else {
+    launch_acc_f32(dst, src, n, stream);
+  }
 }
 
 double reduce_sum_f32(const float* src, size_t n, hipStream_t stream) {

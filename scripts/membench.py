@@ -80,6 +80,8 @@ def main():
     print(f"fill        {t*1e3:9.3f} ms  {n*4/t/1e9:9.1f} GB/s (write)")
     t = time_gpu(lambda: hpk.acc_f32(a.data_ptr(), b.data_ptr(), n, s))
     print(f"accumulate  {t*1e3:9.3f} ms  {3*n*4/t/1e9:9.1f} GB/s (2r+1w)")
+    t = time_gpu(lambda: hpk.acc_f32_nt(a.data_ptr(), b.data_ptr(), n, s))
+    print(f"acc (NT)    {t*1e3:9.3f} ms  {3*n*4/t/1e9:9.1f} GB/s (2r+1w)")
     t = time_gpu(lambda: hpk.reduce_sum_f32(a.data_ptr(), n, s))
     print(f"reduce_sum  {t*1e3:9.3f} ms  {n*4/t/1e9:9.1f} GB/s (read)")
 

@@ -211,3 +211,18 @@ def test_sdma_engine_pair_query(ops, dev):
     # same-device pair: must not crash; >=0 engines
     n = hpk.sdma_num_engines_pair(0, 0)
     assert n >= 0
+
+
+def test_accumulate_nt_variant(ops, dev):
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    n = (48 << 20) // 4  # beyond-L3 class size, 16B aligned
+    a = torch.rand(n, device=dev)
+    b = torch.rand(n, device=dev)
+    ref = a + b
+    torch.cuda.synchronize()
+    hpk.acc_f32_nt(a.data_ptr(), b.data_ptr(), n,
+                   torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert torch.equal(a, ref)
