@@ -38,7 +38,7 @@ def test_iota(ops, dev, n):
     assert torch.equal(t, torch.arange(n, dtype=torch.float32, device=dev))
 
 
-@pytest.mark.parametrize("n", [4, 1023, 1 << 22, (1 << 22) + 3])
+@pytest.mark.parametrize("n", [4, 1023, 1 << 22, (1 << 24) + 3])
 def test_accumulate(ops, dev, n):
     a = torch.rand(n, device=dev)
     b = torch.rand(n, device=dev)
