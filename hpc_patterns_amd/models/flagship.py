@@ -3,13 +3,14 @@
 One "step" per GPU, all submitted concurrently on separate hipStreams
 (reference concurrency pattern), then joined:
 
-  stream 0: K1 busy-wait compute kernel (calibrated to the copy time)
-  stream 1: K2 shader D2D copy (1 GiB payload by default)
-  stream 2: H2D hipMemcpyAsync from pinned host memory
-  stream 3: D2H hipMemcpyAsync to pinned host memory
-  world>1 : all-reduce of 2^25 floats (RCCL) + pairwise 188.7 MB P2P
-            exchange (ncclSend/Recv over xGMI) — the reference's C4/C5/C1
-            communication patterns
+  stream 0:      K1 busy-wait compute kernel (calibrated to the copy time)
+  stream 1:      K2 shader D2D copy, nontemporal (1 GiB payload by default)
+  SDMA engine 0: H2D from pinned host memory (explicit
+                 hsa_amd_memory_async_copy_on_engine — see _h2d)
+  SDMA engine 1: D2H to pinned host memory
+  world>1 :      all-reduce of 2^25 floats (RCCL) + pairwise 188.7 MB P2P
+                 exchange (ncclSend/Recv over xGMI) — the reference's
+                 C4/C5/C1 communication patterns
 
 The benchmark value is whole-job aggregate bandwidth: payload bytes moved by
 all ranks divided by step time (max over ranks). Compute contributes no
