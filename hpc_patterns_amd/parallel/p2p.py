@@ -56,8 +56,11 @@ def pairwise_bandwidth(nbytes: int, iters: int = 10, bidirectional: bool = False
         torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu"))
     send = torch.empty(n, dtype=torch.float32, device=dev)
     recv = torch.empty_like(send)
-    # shuffled-iota checksum payload (reference fill_randomly)
-    perm = torch.randperm(n)
+    # shuffled-iota checksum payload (reference fill_randomly) — seeded per
+    # rank for reproducibility (the reference's default-constructed
+    # minstd_rand had no seed control, SURVEY.md §4 gap)
+    g = torch.Generator().manual_seed(0x5EED + rank)
+    perm = torch.randperm(n, generator=g)
     send.copy_(perm.to(torch.float32))
     expected = float(perm.to(torch.float32).to(torch.float64).sum())
 
