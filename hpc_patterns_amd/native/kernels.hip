@@ -289,11 +289,12 @@ __global__ void k_reduce_partial_f32(const float* __restrict__ src, size_t n,
   double s0 = 0.0, s1 = 0.0, s2 = 0.0, s3 = 0.0;
   size_t stride = (size_t)gridDim.x * blockDim.x;
   size_t n4 = n / 4;
-  const float4* src4 = (const float4*)src;
   if (((uintptr_t)src % 16) == 0) {
+    // nontemporal: the data is read exactly once — don't displace caches
+    const f4_ev_t* src4 = (const f4_ev_t*)src;
     for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
          i += stride) {
-      float4 v = src4[i];
+      f4_ev_t v = __builtin_nontemporal_load(src4 + i);
       s0 += (double)v.x;
       s1 += (double)v.y;
       s2 += (double)v.z;
