@@ -290,11 +290,14 @@ __global__ void k_reduce_partial_f32(const float* __restrict__ src, size_t n,
   size_t stride = (size_t)gridDim.x * blockDim.x;
   size_t n4 = n / 4;
   if (((uintptr_t)src % 16) == 0) {
-    // nontemporal: the data is read exactly once — don't displace caches
-    const f4_ev_t* src4 = (const f4_ev_t*)src;
+    // PLAIN loads on purpose: this kernel is the suite's checksum oracle.
+    // An NT-load variant read a stale/partial view of freshly-copied data
+    // once on gfx950 (p2p checksum exactly half, r24) — verification must
+    // not depend on cache-hint semantics.
+    const float4* src4 = (const float4*)src;
     for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
          i += stride) {
-      f4_ev_t v = __builtin_nontemporal_load(src4 + i);
+      float4 v = src4[i];
       s0 += (double)v.x;
       s1 += (double)v.y;
       s2 += (double)v.z;

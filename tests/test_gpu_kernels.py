@@ -226,3 +226,33 @@ def test_accumulate_nt_variant(ops, dev):
                    torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     assert torch.equal(a, ref)
+
+
+@pytest.mark.parametrize("mb", [40, 64])
+def test_nt_kernels_coherence_stress(ops, dev, mb):
+    """Hunt cache-hint staleness in the NT performance kernels: write fresh
+    data with NORMAL stores (dirty L2), immediately run the NT copy /
+    accumulate on it, verify exactly. Repeated at L2-window sizes — if NT
+    loads/stores bypassed coherence this flakes (cf. the reverted NT
+    reduction, profiles/README r24 note)."""
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    n = (mb << 20) // 4
+    s = torch.cuda.current_stream().cuda_stream
+    for trial in range(4):
+        src = torch.rand(n, device=dev)          # normal stores
+        dst = torch.zeros(n, device=dev)         # normal stores (dirty zeros)
+        torch.cuda.synchronize()
+        ops.copy_kernel(dst, src)                # NT path at these sizes
+        torch.cuda.synchronize()
+        assert torch.equal(dst, src), f"NT copy stale data (trial {trial})"
+
+        a = torch.rand(n, device=dev)
+        b = torch.rand(n, device=dev)
+        ref = a + b
+        torch.cuda.synchronize()
+        hpk.acc_f32_nt(a.data_ptr(), b.data_ptr(), n, s)
+        torch.cuda.synchronize()
+        assert torch.equal(a, ref), f"NT accumulate stale data (trial {trial})"
+        del src, dst, a, b, ref
