@@ -254,6 +254,10 @@ def main() -> int:
         components["p2p_checksum_ok"] = bool(bw["checksum_ok"])
         pp = pingpong(nbytes=8, iters=50, device=device)
         components["pingpong_us"] = round(pp["oneway_us"], 2)
+        # latency->bandwidth ladder (device-buffer ping-pong, one-way µs)
+        for nb, key in ((8 << 10, "pingpong_8k_us"), (8 << 20, "pingpong_8m_us")):
+            ppn = pingpong(nbytes=nb, iters=20, device=device)
+            components[key] = round(ppn["oneway_us"], 2)
         if rank == 0 and not args.cpu:
             xgmi = _measure_xgmi_peer_copy(cfg["p2p_floats"] * 4)
             if xgmi is not None:
