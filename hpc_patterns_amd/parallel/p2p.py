@@ -161,3 +161,51 @@ def pingpong(nbytes: int = 8, iters: int = 100,
         "iters": iters,
         "nbytes": nbytes,
     }
+
+
+def _main(argv=None) -> int:
+    """Latency/bandwidth sweep CLI (BASELINE: "ping-pong µs vs message size"):
+
+        python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \\
+            --master-addr 127.0.0.1 -m hpc_patterns_amd.parallel.p2p
+
+    Prints one-way ping-pong latency from 8 B to 8 MB (doubling) and the
+    pairwise bandwidth phases at the reference 188.7 MB size.
+    """
+    import argparse
+
+    from .init import init_distributed
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--max-bytes", type=int, default=8 << 20)
+    ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--pair-floats", type=int, default=47_185_920)
+    args = ap.parse_args(argv)
+
+    rank, _, world = init_distributed()
+    dev = (torch.device("cuda", torch.cuda.current_device())
+           if torch.cuda.is_available() else torch.device("cpu"))
+    if rank == 0:
+        print(f"# ping-pong one-way latency, world={world}, device={dev.type}")
+        print("bytes,oneway_us")
+    nb = 8
+    while nb <= args.max_bytes:
+        r = pingpong(nbytes=nb, iters=args.iters, device=dev)
+        if rank == 0:
+            print(f"{nb},{r['oneway_us']:.2f}", flush=True)
+        nb *= 2
+    for bidir in (False, True):
+        r = pairwise_bandwidth(args.pair_floats * 4, iters=10,
+                               bidirectional=bidir, device=dev)
+        if rank == 0:
+            d = "bidirectional" if bidir else "unidirectional"
+            print(f"# pairwise {d}: {r['gbps']:.2f} GB/s "
+                  f"(pairs={r['pairs']}, checksum_ok={r['checksum_ok']})")
+    dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+
+    sys.exit(_main())

@@ -47,3 +47,20 @@ def test_bench_single_rank_cpu():
     data = json.loads([l for l in res.stdout.splitlines()
                        if l.startswith("{")][-1])
     assert data["n_gpus"] == 1 and data["value"] > 0
+
+
+def test_p2p_latency_sweep_cli(dist_env):
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        "-m", "hpc_patterns_amd.parallel.p2p",
+        "--max-bytes", "1024", "--iters", "5", "--pair-floats", "4096",
+    ]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "bytes,oneway_us" in res.stdout
+    assert "pairwise unidirectional" in res.stdout
+    assert "checksum_ok=True" in res.stdout
