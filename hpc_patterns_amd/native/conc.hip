@@ -28,6 +28,7 @@
 #include <numeric>
 #include <stdexcept>
 #include <thread>
+#include <future>
 
 namespace hpk {
 
@@ -108,12 +109,19 @@ struct Command {
   int copy_engine = kCopyEngineAuto;
   int sdma_engine = -1; // explicit engine index for kCopyEngineSdma
   mutable std::vector<void*> sdma_handles;
+  mutable std::vector<std::future<void>> staged_futures;
 
-  // SDMA path usable: both sides HIP-registered (not pageable M) and not
-  // inside a graph capture (HSA copies are not capturable).
+  // direct SDMA path usable: both sides HIP-registered (not pageable M) and
+  // not inside a graph capture (HSA copies are not capturable).
   bool sdma_ok() const {
     return !is_compute && copy_engine == kCopyEngineSdma &&
            src.space != 'M' && dst.space != 'M';
+  }
+  // pageable side under the sdma engine: pipelined pinned-staging copy on a
+  // named engine (staged.hip) instead of the runtime's shared staging path.
+  bool staged_ok() const {
+    return !is_compute && copy_engine == kCopyEngineSdma &&
+           (src.space == 'M') != (dst.space == 'M');
   }
 
   void submit(hipStream_t stream, bool in_graph = false) const {
@@ -133,6 +141,17 @@ struct Command {
       (void)hipGetDevice(&dev);
       sdma_handles.push_back(sdma_copy_begin(
           dst.ptr, src.ptr, n_floats * sizeof(float), dev, sdma_engine));
+    } else if (!in_graph && staged_ok()) {
+      int dev = 0;
+      (void)hipGetDevice(&dev);
+      bool h2d = (src.space == 'M');
+      void* d = dst.ptr;
+      const void* s = src.ptr;
+      size_t bytes = n_floats * sizeof(float);
+      int eng = sdma_engine;
+      staged_futures.push_back(std::async(std::launch::async, [=] {
+        staged_copy(d, s, bytes, dev, eng, h2d);
+      }));
     } else {
       check_hip(hipMemcpyAsync(dst.ptr, src.ptr, n_floats * sizeof(float),
                                hipMemcpyDefault, stream),
@@ -143,6 +162,8 @@ struct Command {
   void wait_sdma() const {
     for (void* h : sdma_handles) sdma_wait(h);
     sdma_handles.clear();
+    for (auto& f : staged_futures) f.get();
+    staged_futures.clear();
   }
 };
 
@@ -249,7 +270,7 @@ ConcResult conc_bench(const std::string& mode,
     int nengines = sdma_num_engines(dev);
     int e = 0;
     for (int i = 0; i < ncmds; ++i)
-      if (cmds[i].sdma_ok() && nengines > 0)
+      if ((cmds[i].sdma_ok() || cmds[i].staged_ok()) && nengines > 0)
         cmds[i].sdma_engine = (e++) % nengines;
   }
   // unique streams actually in use (for the end-of-rep sync)

@@ -188,6 +188,16 @@ PYBIND11_MODULE(_hpk, m) {
         },
         py::arg("dst"), py::arg("src"), py::arg("nbytes"), py::arg("device") = 0,
         py::arg("engine_index") = -1);
+  m.def("staged_copy",
+        [](uintptr_t dst, uintptr_t src, size_t nbytes, int device,
+           int engine_index, bool h2d) {
+          py::gil_scoped_release release;
+          hpk::staged_copy(reinterpret_cast<void*>(dst),
+                           reinterpret_cast<const void*>(src), nbytes, device,
+                           engine_index, h2d);
+        },
+        py::arg("dst"), py::arg("src"), py::arg("nbytes"), py::arg("device") = 0,
+        py::arg("engine_index") = 0, py::arg("h2d") = true);
   m.def("sdma_wait", [](uintptr_t handle) {
     py::gil_scoped_release release;
     hpk::sdma_wait(reinterpret_cast<void*>(handle));

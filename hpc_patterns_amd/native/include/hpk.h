@@ -162,6 +162,11 @@ void ipc_close_handle(void* dptr);
 // done natively).
 // ---------------------------------------------------------------------------
 int sdma_num_engines(int device);
+// Pipelined pageable<->device copy through pinned double-buffer staging on
+// a named SDMA engine (blocking call; run it on its own thread to overlap
+// with other commands). h2d: src pageable -> dst device; else reverse.
+void staged_copy(void* dst, const void* src, size_t nbytes, int device,
+                 int engine_index, bool h2d);
 // engines usable for src_device -> dst_device peer copies (xGMI SDMA).
 int sdma_num_engines_pair(int dst_device, int src_device);
 // engine_index < 0 lets ROCr pick. Returns a handle; copy completes when

@@ -256,3 +256,22 @@ def test_nt_kernels_coherence_stress(ops, dev, mb):
         torch.cuda.synchronize()
         assert torch.equal(a, ref), f"NT accumulate stale data (trial {trial})"
         del src, dst, a, b, ref
+
+
+def test_staged_pageable_copy_roundtrip(ops, dev):
+    """Pipelined pinned-staging copies for pageable memory (staged.hip):
+    pageable->device on engine 0, device->pageable on engine 1, exact."""
+    import numpy as np
+
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    n = (40 << 20) // 4 + 13  # 40MB + odd tail, several 8MB chunks
+    src = np.random.rand(n).astype(np.float32)  # pageable numpy memory
+    out = np.zeros_like(src)
+    dbuf = hpk.hip_malloc(n * 4)
+    hpk.staged_copy(dbuf, src.ctypes.data, n * 4, 0, 0, True)
+    hpk.staged_copy(out.ctypes.data, dbuf, n * 4, 0,
+                    1 if hpk.sdma_num_engines(0) >= 2 else 0, False)
+    assert np.array_equal(out, src)
+    hpk.hip_free(dbuf)
