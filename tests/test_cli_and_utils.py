@@ -127,3 +127,23 @@ def test_env_knobs_listing(monkeypatch):
     monkeypatch.setenv("HSA_ENABLE_SDMA", "1")
     monkeypatch.setenv("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     assert check_env() == []
+
+
+def test_native_binary_help_paths():
+    """The native CLIs print usage before any HIP call, so their argument
+    handling is testable without a GPU (skipped if binaries not built)."""
+    conc = REPO / "bin/hpk_conc"
+    if not conc.exists():
+        pytest.skip("binaries not built")
+    res = subprocess.run([str(conc)], capture_output=True, text=True)
+    assert res.returncode == 1 and "Usage" in res.stdout
+    res = subprocess.run([str(conc), "bogus_mode", "--commands", "C"],
+                         capture_output=True, text=True)
+    assert res.returncode == 1 and "unknown mode" in res.stdout
+    res = subprocess.run([str(conc), "in_order", "--commands", "Q2D"],
+                         capture_output=True, text=True)
+    assert res.returncode == 1 and "unsupported COMMAND" in res.stdout
+    for b in ("hpk_allreduce", "hpk_p2p", "hpk_membench"):
+        res = subprocess.run([str(REPO / "bin" / b), "--help"],
+                             capture_output=True, text=True)
+        assert res.returncode == 0 and "Usage" in res.stdout, b
