@@ -14,7 +14,6 @@ from __future__ import annotations
 
 import importlib
 import subprocess
-import sys
 from pathlib import Path
 
 _REPO_ROOT = Path(__file__).resolve().parent.parent
