@@ -211,8 +211,9 @@ class FlagshipPatternStep:
             ar_work.wait()
 
         self._wait_copies()
-        for s in self.streams:
-            s.synchronize()
+        # one device-wide sync instead of four per-stream syncs: measurably
+        # less host overhead per step, identical semantics at step boundary
+        torch.cuda.synchronize()
 
     # ---- accounting ----
     def bytes_per_step_per_rank(self) -> int:
@@ -259,8 +260,7 @@ class FlagshipPatternStep:
             self._h2d(s2)
             self._d2h(s3)
             self._wait_copies()
-            for s in self.streams:
-                s.synchronize()
+            torch.cuda.synchronize()
             return time.perf_counter() - t0
 
         serial_best, per_cmd_best = float("inf"), None
