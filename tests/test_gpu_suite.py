@@ -124,15 +124,16 @@ def test_bin_conc_balanced_pair():
     # 256 MB copies + a real compute grid: at the default 64 MB / 1-thread-C
     # scale the commands finish in tens of µs and overlap is launch-latency
     # noise (observed flaky) — the criterion needs ms-scale commands.
-    # shader copy engine: both commands are our own kernels, so overlap is
-    # co-residency we control (the runtime blit path is box-dependent —
-    # see test_gpu_conc.test_overlap_compute_copy)
+    # C || H2D on a named SDMA engine: kernel and DMA are independent
+    # hardware units, so the reference criterion must pass on every box
+    # (kernel||kernel co-scheduling is box-dependent — see
+    # test_gpu_conc.test_overlap_compute_copy)
     res = _run([str(REPO / "bin/hpk_conc"), "in_order", "--repetitions", "10",
-                "--globalsize_default_memory", str(1 << 26),
-                "--globalsize_C", str(1 << 16), "--copy_engine", "shader",
-                "--commands", "C", "D2D"])
-    assert "## in_order | C DD |" in res.stdout, res.stdout + res.stderr
-    # the balanced C||D2D overlap must actually pass the reference criterion
+                "--globalsize_HD", str(1 << 26),
+                "--globalsize_C", str(1 << 16), "--copy_engine", "sdma",
+                "--commands", "C", "H2D"])
+    assert "## in_order | C HD |" in res.stdout, res.stdout + res.stderr
+    # the balanced C||H2D overlap must pass the reference criterion
     assert "SUCCESS" in res.stdout, res.stdout
 
 
