@@ -76,8 +76,27 @@ double fill_payload(float* dptr, size_t n, unsigned seed) {
 void verify(float* dptr, size_t n, double expected, const char* what) {
   double got = hpk::reduce_sum_f32(dptr, n, nullptr);
   if (got != expected) {
-    std::fprintf(stderr, "CHECKSUM FAILURE (%s): got %.1f expected %.1f\n",
-                 what, got, expected);
+    // Second opinion before declaring failure: full device sync + host-side
+    // sum distinguishes "buffer really wrong" from "device reduction raced
+    // something" (an intermittent ~half-sum was observed on some boxes —
+    // see profiles/README.md).
+    hpk::check_hip(hipDeviceSynchronize(), "verify sync");
+    std::vector<float> h(n);
+    hpk::check_hip(hipMemcpy(h.data(), dptr, n * sizeof(float),
+                             hipMemcpyDeviceToHost),
+                   "verify D2H");
+    double host_sum = 0.0;
+    for (size_t i = 0; i < n; ++i) host_sum += (double)h[i];
+    double got2 = hpk::reduce_sum_f32(dptr, n, nullptr);
+    std::fprintf(stderr,
+                 "CHECKSUM FAILURE (%s): device %.1f, device-after-sync %.1f, "
+                 "host %.1f, expected %.1f\n",
+                 what, got, got2, host_sum, expected);
+    if (host_sum == expected && got2 == expected) {
+      std::fprintf(stderr,
+                   "  -> buffer correct after device sync: first reduction "
+                   "raced the copy completion\n");
+    }
     std::exit(2);
   }
 }
