@@ -497,6 +497,7 @@ void launch_acc_i32(int* dst, const int* src, size_t n, hipStream_t stream) {
 }
 
 long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream) {
+  check_hip(hipDeviceSynchronize(), "reduce_sum_i32 pre-sync"); // see f32
   constexpr size_t kRedBlocks = 4096;
   size_t blocks = (n + (size_t)kBlock * 8 - 1) / ((size_t)kBlock * 8);
   if (blocks == 0) blocks = 1;
@@ -538,6 +539,13 @@ void launch_acc_f32_nt(float* dst, const float* src, size_t n,
 }
 
 double reduce_sum_f32(const float* src, size_t n, hipStream_t stream) {
+  // This is the suite's checksum oracle: device-wide sync first. Measured
+  // on ROCm 7.2 (profiles/p2p_fail_r37.log): a kernel launched on another
+  // stream immediately after hipStreamSynchronize() of the producing
+  // stream intermittently (~1/25) read a partially-visible buffer; the
+  // data was correct after hipDeviceSynchronize(). Verification must not
+  // depend on cross-stream visibility timing.
+  check_hip(hipDeviceSynchronize(), "reduce_sum_f32 pre-sync");
   constexpr size_t kRedBlocks = 4096;
   size_t blocks = (n + (size_t)kBlock * 8 - 1) / ((size_t)kBlock * 8);
   if (blocks == 0) blocks = 1;
