@@ -81,7 +81,11 @@ clean-bins:
 test:
 	$(PYTHON) -m pytest tests -q -m "not gpu"
 
+# full GPU battery (tests + performance floors + miniapps) — needs an MI355X
+gpu-test:
+	bash scripts/ci_gpu.sh
+
 clean:
 	rm -rf $(BUILD) $(BIN) $(EXT_SO)
 
-.PHONY: asan clean-bins test
+.PHONY: asan clean-bins test gpu-test

@@ -140,3 +140,17 @@ def test_format_time_info():
 def test_verdict_line_grammar():
     line = verdict_line("in_order", ["C", "MD"], "SUCCESS: x")
     assert line == "## in_order | C MD | SUCCESS: x"
+
+
+def test_run_bench_rejects_bad_copy_engine():
+    from hpc_patterns_amd.concurrency import run_bench
+
+    with pytest.raises(ValueError, match="copy_engine"):
+        run_bench("serial", ["C"], copy_engine="warp_drive")
+
+
+def test_run_bench_rejects_bad_mode_before_native():
+    from hpc_patterns_amd.concurrency import run_bench
+
+    with pytest.raises(ValueError, match="mode"):
+        run_bench("hyperspeed", ["C"])

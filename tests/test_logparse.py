@@ -39,3 +39,13 @@ def test_reference_style_log_lines():
     ]
     parsed = parse_log(lines)
     assert parsed["ZE_AFFINITY_MASK=0.0"]["C C"]["out_of_order"] == "SUCCESS"
+
+
+def test_parse_tolerates_malformed_verdict_lines():
+    lines = [
+        "export X=1",
+        "## only-one-field SUCCESS",       # no | separator -> skipped
+        "## a | b | SUCCESS: ok",
+    ]
+    parsed = parse_log(lines)
+    assert parsed["X=1"] == {"b": {"a": "SUCCESS"}}
