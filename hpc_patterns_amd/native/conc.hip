@@ -226,14 +226,35 @@ ConcResult conc_bench(const std::string& mode,
   // copies, device<->device copies — and copy streams never see a kernel.
   static std::vector<hipStream_t> pool_kernel, pool_hostcopy, pool_devcopy;
   static hipStream_t master_stream = nullptr;
-  auto take = [&](std::vector<hipStream_t>& pool, size_t idx) {
+  // HPK_COPY_STREAM_PRIORITY=high raises the copy pools' stream priority:
+  // on pods where two saturating kernels co-schedule poorly (1.3-1.4x,
+  // profiles/README.md), a higher-priority copy stream lets copy-kernel
+  // blocks interleave with the compute kernel's.
+  static const bool hi_prio_copies = [] {
+    const char* env = std::getenv("HPK_COPY_STREAM_PRIORITY");
+    return env && std::string(env) == "high";
+  }();
+  auto take_prio = [&](std::vector<hipStream_t>& pool, size_t idx, bool hi) {
     while (pool.size() <= idx) {
       hipStream_t s;
-      check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking),
-                "stream create");
+      if (hi) {
+        int least = 0, greatest = 0;
+        check_hip(hipDeviceGetStreamPriorityRange(&least, &greatest),
+                  "priority range");
+        check_hip(hipStreamCreateWithPriority(&s, hipStreamNonBlocking,
+                                              greatest),
+                  "stream create (prio)");
+      } else {
+        check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking),
+                  "stream create");
+      }
       pool.push_back(s);
     }
     return pool[idx];
+  };
+  auto take = [&](std::vector<hipStream_t>& pool, size_t idx) {
+    bool hi = hi_prio_copies && (&pool != &pool_kernel);
+    return take_prio(pool, idx, hi);
   };
   if (master_stream == nullptr)
     check_hip(hipStreamCreateWithFlags(&master_stream, hipStreamNonBlocking),
