@@ -31,6 +31,8 @@
 //   -n <N>     number of ranks (default: visible device count)
 //   -i <iters> timed iterations (default 10, min-time reported)
 //   -c <K>     pipeline chunk count (default 8; pipeline algo only)
+//   -t <T>     dtype: float (default) | int — the reference's two
+//              -DAPP_DATA_TYPE instantiations as a runtime switch
 //   --algo ring|pipeline|rccl
 
 #include "../hpc_patterns_amd/native/include/hpk.h"
