@@ -64,3 +64,21 @@ def test_p2p_latency_sweep_cli(dist_env):
     assert "bytes,oneway_us" in res.stdout
     assert "pairwise unidirectional" in res.stdout
     assert "checksum_ok=True" in res.stdout
+
+
+def test_allreduce_sweep_cli(dist_env):
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        "-m", "hpc_patterns_amd.parallel.sweep",
+        "--min-mb", "0.01", "--max-mb", "0.02", "--iters", "2",
+        "--algos", "rccl,ring,rsag",
+    ]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert "algo,bytes,time_s,alg_GBps,bus_GBps" in res.stdout
+    for algo in ("rccl", "ring", "rsag"):
+        assert f"\n{algo}," in res.stdout or res.stdout.startswith(f"{algo},"), algo
