@@ -82,3 +82,23 @@ def test_allreduce_sweep_cli(dist_env):
     assert "algo,bytes,time_s,alg_GBps,bus_GBps" in res.stdout
     for algo in ("rccl", "ring", "rsag"):
         assert f"\n{algo}," in res.stdout or res.stdout.startswith(f"{algo},"), algo
+
+
+def test_bench_torchrun_cpu_world8(dist_env):
+    """The driver's N=8 launch shape, rehearsed end to end on gloo."""
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "8",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        str(REPO / "bench.py"), "--gpus", "8", "--steps", "2",
+        "--warmup", "1", "--cpu",
+    ]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    data = json.loads([l for l in res.stdout.splitlines()
+                       if l.startswith("{")][0])
+    assert data["n_gpus"] == 8
+    assert data["config"]["parallelism"] == "dp8"
+    assert data["components"]["p2p_checksum_ok"] is True
