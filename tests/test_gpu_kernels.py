@@ -275,3 +275,23 @@ def test_staged_pageable_copy_roundtrip(ops, dev):
                     1 if hpk.sdma_num_engines(0) >= 2 else 0, False)
     assert np.array_equal(out, src)
     hpk.hip_free(dbuf)
+
+
+@pytest.mark.parametrize("nbytes", [1024, 8 << 20, (8 << 20) + 4,
+                                    (16 << 20) - 4, 65 << 20])
+def test_staged_copy_chunk_boundaries(ops, dev, nbytes):
+    """staged.hip chunk/thread-split logic at boundary sizes (8 MiB chunks,
+    4-way threading beyond 64 MiB)."""
+    import numpy as np
+
+    from hpc_patterns_amd._native import native
+
+    hpk = native()
+    n = nbytes // 4
+    src = np.random.rand(n).astype(np.float32)
+    out = np.zeros_like(src)
+    dbuf = hpk.hip_malloc(nbytes)
+    hpk.staged_copy(dbuf, src.ctypes.data, nbytes, 0, 0, True)
+    hpk.staged_copy(out.ctypes.data, dbuf, nbytes, 0, 0, False)
+    assert np.array_equal(out, src), nbytes
+    hpk.hip_free(dbuf)
