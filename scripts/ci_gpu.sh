@@ -31,6 +31,11 @@ echo "== miniapps"
 ./bin/hpk_interop || rc=1
 ./bin/hpk_membench --quick || rc=1
 
+echo "== cmake + ctest harness (the reference's build contract)"
+rm -rf build-cmake && mkdir -p build-cmake
+(cd build-cmake && CXX=/opt/rocm/bin/hipcc cmake .. > cmake.log 2>&1 \
+   && make -j16 > make.log 2>&1 && ctest --output-on-failure) || rc=1
+
 echo "== bench smoke"
 python bench.py --smoke --steps 3 --warmup 1 || rc=1
 
