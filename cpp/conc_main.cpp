@@ -241,8 +241,7 @@ int main(int argc, char* argv[]) {
                 << time_info({cmds[i]}, serial.per_cmd_us[i], params, -1,
                              nullptr)
                 << std::endl;
-      if (enable_profiling &&
-          serial.per_cmd_dev_ms[i] != std::numeric_limits<double>::max())
+      if (enable_profiling && serial.per_cmd_dev_ms[i] >= 0.0)
         std::cout << "    Device Time (hipEvent): " << serial.per_cmd_dev_ms[i]
                   << "ms" << std::endl;
     }
@@ -260,6 +259,14 @@ int main(int argc, char* argv[]) {
     std::string conc_info =
         time_info(cmds, conc.total_us, params, min_bandwidth, &bw_errno);
     std::cout << "Minimum Measured Total Time //: " << conc_info << std::endl;
+    if (enable_profiling) {
+      // per-command device times in the CONCURRENT run (graph modes report
+      // these via event-record graph nodes — r1 left them unmeasured)
+      for (size_t i = 0; i < cmds.size(); ++i)
+        if (conc.per_cmd_dev_ms[i] >= 0.0)
+          std::cout << "  // Device Time Command " << i << " (" << cmds[i]
+                    << "): " << conc.per_cmd_dev_ms[i] << "ms" << std::endl;
+    }
     double speedup = (double)serial.total_us / (double)std::max(conc.total_us, 1L);
     std::cout << "Speedup Relative to Serial: " << speedup << "x" << std::endl;
 

@@ -20,7 +20,7 @@ from __future__ import annotations
 from typing import Mapping, Sequence
 
 ALLOWED_MODES = ("serial", "in_order", "out_of_order", "graph",
-                 "host_threads", "nowait")
+                 "graph_explicit", "host_threads", "nowait")
 
 DEFAULT_TRIPCOUNT = 40_000
 DEFAULT_GLOBALSIZE_C = 1

@@ -87,6 +87,9 @@ long long reduce_sum_i32(const int* src, size_t n, hipStream_t stream);
 //                  HIP analog of a SYCL out-of-order queue
 //   out_of_order — alias of graph (HIP streams are strictly in-order; the
 //                  graph scheduler is the runtime-managed concurrency path)
+//   graph_explicit — same graph semantics but built with explicit node-API
+//                  calls (hipGraphAddMemcpyNode1D / child-graph kernel
+//                  nodes, all independent roots) instead of stream capture
 //   nowait       — alias of in_order (reference bench_omp.cpp nowait mode)
 extern const std::string allowed_modes;
 bool mode_is_allowed(const std::string& mode);
@@ -103,6 +106,8 @@ struct ConcResult {
   long total_us = 0;                  // min over repetitions
   std::vector<long> per_cmd_us;       // serial mode: min per-command wall time
   std::vector<double> per_cmd_dev_ms; // hipEvent device time (profiling only)
+  // unmeasured per-command entries are -1 (e.g. per_cmd_us outside serial
+  // mode, per_cmd_dev_ms without --enable_profiling)
 };
 
 // commands: "C" or "A2B"/"AB" with A,B in {M,D,H,S} =

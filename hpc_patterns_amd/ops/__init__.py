@@ -56,6 +56,9 @@ def copy_kernel(dst: torch.Tensor, src: torch.Tensor, stream=None) -> None:
         raise ValueError("size mismatch")
     if not (dst.is_contiguous() and src.is_contiguous()):
         raise TypeError("contiguous tensors required")
+    if not (dst.is_cuda and src.is_cuda):
+        raise TypeError("dst and src must be CUDA tensors (a host pointer "
+                        "would crash the GPU copy kernel)")
     native().copy_kernel(dst.data_ptr(), src.data_ptr(),
                          dst.numel() * dst.element_size(),
                          _stream_handle(stream))
