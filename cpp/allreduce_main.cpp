@@ -74,6 +74,7 @@ struct Config {
   char alloc = 'D';
   std::string algo = "ring";
   std::string dtype = "float"; // float | int (reference -DAPP_DATA_TYPE pair)
+  std::string transport = "nccl"; // nccl | ipc (ring exchange transport)
   int nranks = -1;
   int iters = 10;
   int chunks = 8;
@@ -121,29 +122,110 @@ void* alloc_buf(char kind, size_t bytes) {
   return p;
 }
 
+// Ring exchange transports. The ring ALGORITHM below is transport-agnostic;
+// these provide one step's neighbour exchange:
+//   NcclRing — grouped ncclSend/ncclRecv, fully async on `stream` (the RCCL
+//              pt2pt path; no odd/even ordering dance — that deadlock-
+//              avoidance trick is an MPI-blocking-call artifact, reference
+//              allreduce-mpi-sycl.cpp:50-58)
+//   IpcRing  — one-sided put into the RIGHT neighbour's recv buffer through
+//              a hipIpc mapping + shared-memory fence. RCCL refuses two
+//              ranks on one device, so this is what lets the ring run
+//              oversubscribed (size>=2 on a 1-GPU box) under ctest, the way
+//              the reference's `mpirun -np 4` oversubscribed tiles
+//              (CMakeLists.txt:45-50).
+template <typename T>
+struct NcclRing {
+  ncclComm_t comm;
+  int right, left;
+  void sendrecv(T* send, T* recv, size_t n, hipStream_t stream) {
+    const ncclDataType_t dt = hpk::get_rccl_datatype<T>();
+    check_nccl(ncclGroupStart(), "group start");
+    check_nccl(ncclSend(send, n, dt, right, comm, stream), "send");
+    check_nccl(ncclRecv(recv, n, dt, left, comm, stream), "recv");
+    check_nccl(ncclGroupEnd(), "group end");
+  }
+};
+
+constexpr int kMaxArRanks = 16;
+
+struct ArIpcShared {
+  hpk_launch::SharedBarrier bar;
+  // each rank exports BOTH ring buffers (va/vb swap identity every step,
+  // all ranks in lockstep, so buffer index i on my side pairs with buffer
+  // index i on every neighbour)
+  uint8_t handle[kMaxArRanks][2][sizeof(hipIpcMemHandle_t)];
+  double dt[kMaxArRanks]; // wall-time MAX-reduction without a communicator
+};
+
+template <typename T>
+struct IpcRing {
+  T* alloc0 = nullptr; // my two exchange buffers (allocation identity)
+  T* alloc1 = nullptr;
+  void* right_buf[2] = {nullptr, nullptr}; // right neighbour's, IPC-mapped
+  bool same_dev = false;
+  ArIpcShared* sh = nullptr;
+  bool ok = true;
+
+  void setup(int rank, int size, int ndev, ArIpcShared* shared, T* a0, T* a1) {
+    sh = shared;
+    alloc0 = a0;
+    alloc1 = a1;
+    int right = (rank + 1) % size;
+    same_dev = (rank % ndev) == (right % ndev);
+    auto h0 = hpk::ipc_get_handle(a0);
+    auto h1 = hpk::ipc_get_handle(a1);
+    std::memcpy(sh->handle[rank][0], h0.data(), h0.size());
+    std::memcpy(sh->handle[rank][1], h1.data(), h1.size());
+    ok = sh->bar.wait();
+    if (!ok) return;
+    for (int i = 0; i < 2; ++i) {
+      std::vector<uint8_t> hv(sh->handle[right][i],
+                              sh->handle[right][i] + sizeof(hipIpcMemHandle_t));
+      right_buf[i] = hpk::ipc_open_handle(hv);
+    }
+  }
+
+  void sendrecv(T* send, T* recv, size_t n, hipStream_t stream) {
+    // my recv allocation index == right's recv allocation index (lockstep)
+    int idx = (recv == alloc1) ? 1 : 0;
+    void* dst = right_buf[idx];
+    // same-device put: hand-written copy kernel (release-semantics
+    // completion, robust against the same-device SDMA visibility artifact);
+    // cross-device: SDMA over xGMI
+    if (same_dev)
+      hpk::launch_copy_kernel(dst, send, n * sizeof(T), stream);
+    else
+      hpk::check_hip(hipMemcpyAsync(dst, send, n * sizeof(T),
+                                    hipMemcpyDeviceToDevice, stream),
+                     "ipc ring put");
+    hpk::check_hip(hipStreamSynchronize(stream), "ipc ring sync");
+    ok = ok && sh->bar.wait(); // fence: every put landed -> recv is valid
+  }
+
+  void teardown() {
+    for (int i = 0; i < 2; ++i)
+      if (right_buf[i]) hpk::ipc_close_handle(right_buf[i]);
+  }
+};
+
 // Hand ring all-reduce, the reference SendRecvRing pattern: (size-1) steps of
 // [exchange full buffer with ring neighbours] + [VC += recv]. Buffers VA
 // (send payload, swapped with VB each step), VB (recv), VC (accumulator).
-template <typename T>
-double run_ring(ncclComm_t comm, hipStream_t stream, T* va, T* vb,
-                T* vc, size_t n, int rank, int size) {
-  const ncclDataType_t dt = hpk::get_rccl_datatype<T>();
-  int right = (rank + 1) % size;
-  int left = (rank - 1 + size) % size;
+template <typename T, typename Ring>
+double run_ring(Ring& ring, hipStream_t stream, T* va, T* vb,
+                T* vc, size_t n, int size, int* steps_out = nullptr) {
   double t0 = now_s();
+  int steps = 0;
   Kern<T>::acc(vc, va, n, stream); // VC += own VA (VC starts at 0)
   for (int step = 0; step < size - 1; ++step) {
-    // RCCL pt2pt: group makes the send+recv concurrent (no odd/even ordering
-    // dance needed — that deadlock-avoidance trick is an MPI-blocking-call
-    // artifact, reference allreduce-mpi-sycl.cpp:50-58).
-    check_nccl(ncclGroupStart(), "group start");
-    check_nccl(ncclSend(va, n, dt, right, comm, stream), "send");
-    check_nccl(ncclRecv(vb, n, dt, left, comm, stream), "recv");
-    check_nccl(ncclGroupEnd(), "group end");
+    ring.sendrecv(va, vb, n, stream);
     Kern<T>::acc(vc, vb, n, stream);
     std::swap(va, vb);
+    ++steps;
   }
   hpk::check_hip(hipStreamSynchronize(stream), "ring sync");
+  if (steps_out) *steps_out = steps;
   return now_s() - t0;
 }
 
@@ -197,13 +279,16 @@ double run_pipeline(ncclComm_t comm, hipStream_t comm_stream,
 
 template <typename T>
 int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
-           const Config& cfg) {
+           ArIpcShared* ipc_sh, const Config& cfg) {
+  const bool use_ipc = (cfg.transport == "ipc");
   int dev = rank % ndev;
   hpk::check_hip(hipSetDevice(dev), "hipSetDevice");
-  ncclUniqueId id;
-  hpk_launch::bootstrap_id(sh, rank, &id);
-  ncclComm_t comm;
-  check_nccl(ncclCommInitRank(&comm, size, id, rank), "ncclCommInitRank");
+  ncclComm_t comm = nullptr;
+  if (!use_ipc) {
+    ncclUniqueId id;
+    hpk_launch::bootstrap_id(sh, rank, &id);
+    check_nccl(ncclCommInitRank(&comm, size, id, rank), "ncclCommInitRank");
+  }
 
   size_t n = 1ull << cfg.p;
   size_t bytes = n * sizeof(T);
@@ -216,13 +301,26 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
   hpk::check_hip(hipStreamCreateWithFlags(&comp_stream, hipStreamNonBlocking),
                  "s2");
 
+  int right = (rank + 1) % size;
+  int left = (rank - 1 + size) % size;
+  NcclRing<T> nring{comm, right, left};
+  IpcRing<T> iring;
+  if (use_ipc) {
+    iring.setup(rank, size, ndev, ipc_sh, va, vb);
+    if (!iring.ok) return 3;
+  }
+
   double best = 1e30;
+  int ring_steps = 0;
   for (int it = 0; it < cfg.iters; ++it) {
     // (re)initialize: VA = rank, VB = -1, VC = 0 (reference Initialize)
     Kern<T>::fill(va, (T)rank, n, stream);
     Kern<T>::fill(vb, (T)-1, n, stream);
     Kern<T>::fill(vc, (T)0, n, stream);
     hpk::check_hip(hipStreamSynchronize(stream), "init sync");
+    // ipc transport: nobody may put into my freshly-filled recv buffer
+    // before the fill above completed everywhere
+    if (use_ipc && !ipc_sh->bar.wait()) return 3;
 
     double dt;
     if (cfg.algo == "rccl") {
@@ -235,24 +333,33 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
     } else if (cfg.algo == "pipeline") {
       dt = run_pipeline(comm, stream, comp_stream, va, vb, vc, n, rank, size,
                         cfg.chunks);
+    } else if (use_ipc) {
+      dt = run_ring(iring, stream, va, vb, vc, n, size, &ring_steps);
+      if (!iring.ok) return 3;
     } else {
-      dt = run_ring(comm, stream, va, vb, vc, n, rank, size);
+      dt = run_ring(nring, stream, va, vb, vc, n, size, &ring_steps);
     }
     best = std::min(best, dt);
   }
 
   // max over ranks (reference MPI_Allreduce MAX of wall time)
-  double* d_time = nullptr;
-  hpk::check_hip(hipMalloc(&d_time, sizeof(double)), "time buf");
-  hpk::check_hip(hipMemcpy(d_time, &best, sizeof(double), hipMemcpyHostToDevice),
-                 "time h2d");
-  check_nccl(ncclAllReduce(d_time, d_time, 1, ncclDouble, ncclMax, comm, stream),
-             "time max");
-  hpk::check_hip(hipStreamSynchronize(stream), "time sync");
-  double max_time;
-  hpk::check_hip(hipMemcpy(&max_time, d_time, sizeof(double),
-                           hipMemcpyDeviceToHost),
-                 "time d2h");
+  double max_time = best;
+  if (use_ipc) {
+    ipc_sh->dt[rank] = best;
+    if (!ipc_sh->bar.wait()) return 3;
+    for (int r = 0; r < size; ++r) max_time = std::max(max_time, ipc_sh->dt[r]);
+  } else {
+    double* d_time = nullptr;
+    hpk::check_hip(hipMalloc(&d_time, sizeof(double)), "time buf");
+    hpk::check_hip(hipMemcpy(d_time, &best, sizeof(double),
+                             hipMemcpyHostToDevice), "time h2d");
+    check_nccl(ncclAllReduce(d_time, d_time, 1, ncclDouble, ncclMax, comm,
+                             stream), "time max");
+    hpk::check_hip(hipStreamSynchronize(stream), "time sync");
+    hpk::check_hip(hipMemcpy(&max_time, d_time, sizeof(double),
+                             hipMemcpyDeviceToHost), "time d2h");
+    (void)hipFree(d_time);
+  }
 
   // analytic verification: every element == size*(size-1)/2
   double expected = (double)n * ((double)size * (size - 1) / 2.0);
@@ -266,15 +373,15 @@ int worker(int rank, int size, int ndev, hpk_launch::SharedBootstrap* sh,
     // bus bandwidth convention: ring moves 2(size-1)/size * bytes per rank
     double busbw =
         size > 1 ? 2.0 * (size - 1) / size * gb / max_time : gb / max_time;
-    std::printf("# algo=%s dtype=%s ranks=%d elems=2^%d alloc=%c "
-                "time=%.6fs busbw=%.2f GB/s\n",
-                cfg.algo.c_str(), cfg.dtype.c_str(), size, cfg.p, cfg.alloc,
-                max_time, busbw);
+    std::printf("# algo=%s transport=%s dtype=%s ranks=%d elems=2^%d alloc=%c "
+                "steps=%d time=%.6fs busbw=%.2f GB/s\n",
+                cfg.algo.c_str(), cfg.transport.c_str(), cfg.dtype.c_str(),
+                size, cfg.p, cfg.alloc, ring_steps, max_time, busbw);
     if (sh) sh->result = max_time;
   }
 
-  ncclCommDestroy(comm);
-  (void)hipFree(d_time);
+  if (use_ipc) iring.teardown();
+  if (comm) ncclCommDestroy(comm);
   return pass ? 0 : 2;
 }
 
@@ -301,16 +408,33 @@ int main(int argc, char* argv[]) {
     else if (s == "-c") cfg.chunks = std::atoi(next());
     else if (s == "--algo") cfg.algo = next();
     else if (s == "-t" || s == "--dtype") cfg.dtype = next();
+    else if (s == "--transport") cfg.transport = next();
     else {
       std::printf(
           "Usage: %s [-p P] [-D|-H|-S] [-a] [-n ranks] [-i iters] [-c chunks] "
-          "[--algo ring|pipeline|rccl]\n", argv[0]);
+          "[--algo ring|pipeline|rccl] [--transport nccl|ipc]\n", argv[0]);
       return s == "-h" || s == "--help" ? 0 : 1;
     }
   }
   if (cfg.algo != "ring" && cfg.algo != "pipeline" && cfg.algo != "rccl") {
     std::fprintf(stderr, "unknown algo '%s'\n", cfg.algo.c_str());
     return 1;
+  }
+  if (cfg.transport != "nccl" && cfg.transport != "ipc") {
+    std::fprintf(stderr, "unknown transport '%s'\n", cfg.transport.c_str());
+    return 1;
+  }
+  if (cfg.transport == "ipc") {
+    if (cfg.algo != "ring") {
+      std::fprintf(stderr, "--transport ipc supports --algo ring only "
+                   "(the pipeline/collective paths are RCCL)\n");
+      return 1;
+    }
+    if (cfg.alloc != 'D') {
+      std::fprintf(stderr, "--transport ipc requires -D (hipIpc shares "
+                   "hipMalloc memory only)\n");
+      return 1;
+    }
   }
 
   // Parent stays HIP-free (launch_util.h invariant): probe device count by
@@ -321,18 +445,32 @@ int main(int argc, char* argv[]) {
     return 1;
   }
   int size = cfg.nranks > 0 ? cfg.nranks : ndev;
-  if (size > ndev) {
-    // RCCL refuses two ranks on one device; clamp with a notice instead of
-    // oversubscribing (the reference oversubscribed tiles; xGMI pt2pt
-    // between co-located ranks would be a self-copy anyway).
-    std::fprintf(stderr, "# clamping ranks %d -> %d (one rank per GPU)\n",
-                 size, ndev);
+  if (cfg.transport == "ipc") {
+    // ipc transport oversubscribes freely: many ranks per GPU is exactly
+    // its purpose (multi-rank ring exchange under ctest on a 1-GPU lease)
+    if (size < 2) size = 2;
+    if (size > kMaxArRanks) size = kMaxArRanks;
+  } else if (size > ndev) {
+    // RCCL refuses two ranks on one device; clamp with a notice and point
+    // at the transport that CAN oversubscribe.
+    std::fprintf(stderr, "# clamping ranks %d -> %d (one rank per GPU; "
+                 "use --transport ipc to oversubscribe)\n", size, ndev);
     size = ndev;
   }
 
   hpk_launch::SharedBootstrap* sh = hpk_launch::map_shared();
+  ArIpcShared* ipc_sh = nullptr;
+  if (cfg.transport == "ipc") {
+    ipc_sh = hpk_launch::map_shared_struct<ArIpcShared>();
+    if (!ipc_sh) {
+      std::fprintf(stderr, "shared map failed\n");
+      return 1;
+    }
+    ipc_sh->bar.size = size;
+  }
   return hpk_launch::fork_workers(size, [&](int rank) {
-    return cfg.dtype == "int" ? worker<int>(rank, size, ndev, sh, cfg)
-                              : worker<float>(rank, size, ndev, sh, cfg);
+    return cfg.dtype == "int"
+               ? worker<int>(rank, size, ndev, sh, ipc_sh, cfg)
+               : worker<float>(rank, size, ndev, sh, ipc_sh, cfg);
   });
 }

@@ -192,81 +192,127 @@ int run_peer(size_t n, bool bidir_phase) {
 }
 
 // ---------------------------------------------------------------------------
-// Engine 2: HIP-IPC one-sided put between two processes.
+// Engine 2: HIP-IPC one-sided RMA — full reference protocol parity
+// (reference p2p/peer2pear.cpp:68-102,119-122,141-155): every rank exposes a
+// window (hipIpcMemHandle = the MPI_Win_create), all N/2 pairs run
+// concurrently, a shared-memory barrier is the fence epoch, phase 1 is a
+// unidirectional put (even rank -> odd peer's window), phase 2 bidirectional
+// (both directions), aggregate GB/s = bytes x pairs x dirs / min global
+// interval over 10 iterations, output isomorphic to the two-sided engines.
+// On a 1-GPU box the ranks oversubscribe the device (cross-PROCESS one-sided
+// puts still exercise the whole protocol — the reference oversubscribed
+// tiles the same way, CMakeLists.txt:45-50).
 // ---------------------------------------------------------------------------
-int run_ipc(size_t n) {
-  // IMPORTANT: fork BEFORE any HIP call — an initialized HIP runtime does
-  // not survive fork(), so each side initializes its own.
+constexpr int kMaxIpcRanks = 16;
+
+struct IpcShared {
+  hpk_launch::SharedBarrier bar;
+  uint8_t handle[kMaxIpcRanks][sizeof(hipIpcMemHandle_t)];
+  double src_sum[kMaxIpcRanks]; // expected checksum of each rank's payload
+  double dt[kMaxIpcRanks];      // per-rank interval for the clock union
+};
+
+int ipc_rma_worker(int rank, int size, int ndev, IpcShared* sh, size_t n) {
+  int dev = rank % std::max(ndev, 1);
+  hpk::check_hip(hipSetDevice(dev), "set dev");
   size_t bytes = n * sizeof(float);
-  int sv[2];
-  if (socketpair(AF_UNIX, SOCK_STREAM, 0, sv) != 0) {
-    perror("socketpair");
-    return 1;
+
+  // window (exposed) + source payload (origin side)
+  float *win = nullptr, *src = nullptr;
+  hpk::check_hip(hipMalloc(&win, bytes), "win");
+  hpk::check_hip(hipMalloc(&src, bytes), "src");
+  auto hv = hpk::ipc_get_handle(win);
+  std::memcpy(sh->handle[rank], hv.data(), hv.size());
+  sh->src_sum[rank] = fill_payload(src, n, (unsigned)rank);
+  if (!sh->bar.wait()) return 3; // handle-exchange epoch
+
+  int peer = (rank % 2 == 0) ? rank + 1 : rank - 1;
+  bool paired = peer < size;
+  float* peer_win = nullptr;
+  int peer_dev = paired ? peer % std::max(ndev, 1) : dev;
+  if (paired) {
+    std::vector<uint8_t> ph(sh->handle[peer],
+                            sh->handle[peer] + sizeof(hipIpcMemHandle_t));
+    peer_win = (float*)hpk::ipc_open_handle(ph);
   }
+  hipStream_t s;
+  hpk::check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking), "s");
 
-  pid_t pid = fork();
-  if (pid == 0) {
-    // child = rank 1 (the "origin" of the one-sided put)
-    close(sv[0]);
-    int ndev = hpk::device_count();
-    int dev = ndev >= 2 ? 1 : 0;
-    hpk::check_hip(hipSetDevice(dev), "child set dev");
-    // receive target handle
-    hipIpcMemHandle_t h;
-    if (read(sv[1], &h, sizeof(h)) != sizeof(h)) std::exit(1);
-    std::vector<uint8_t> hv((uint8_t*)&h, (uint8_t*)&h + sizeof(h));
-    void* target = hpk::ipc_open_handle(hv);
-
-    float* src = nullptr;
-    hpk::check_hip(hipMalloc(&src, bytes), "child src");
-    double sum = fill_payload(src, n, 1234);
-    // send the expected checksum (the "window" metadata)
-    if (write(sv[1], &sum, sizeof(sum)) != sizeof(sum)) std::exit(1);
-
-    hipStream_t s;
-    hpk::check_hip(hipStreamCreateWithFlags(&s, hipStreamNonBlocking), "s");
+  int rc = 0;
+  for (int phase = 0; phase < 2; ++phase) {
+    bool bidir = (phase == 1);
+    bool origin = paired && ((rank % 2 == 0) || bidir);
     double best = 1e30;
     for (int it = 0; it < kIters; ++it) {
-      char token;
-      if (read(sv[1], &token, 1) != 1) std::exit(1); // fence: target ready
+      if (!sh->bar.wait()) return 3; // fence: epoch open
       double t0 = now_s();
-      // one-sided put: direct write into the peer process's device buffer
-      hpk::check_hip(hipMemcpyAsync(target, src, bytes, hipMemcpyDeviceToDevice,
-                                    s),
-                     "ipc put");
-      hpk::check_hip(hipStreamSynchronize(s), "ipc sync");
-      double dt = now_s() - t0;
-      best = std::min(best, dt);
-      if (write(sv[1], &dt, sizeof(dt)) != sizeof(dt)) std::exit(1); // fence
+      if (origin) {
+        // one-sided put into the peer process's window. Cross-device: SDMA
+        // over xGMI (hipMemcpyAsync on the IPC mapping). Same device
+        // (oversubscribed 1-GPU box): the hand-written copy kernel — its
+        // completion signal carries release semantics, dodging the SDMA
+        // same-device completion-visibility artifact (profiles/README.md).
+        if (peer_dev == dev)
+          hpk::launch_copy_kernel(peer_win, src, bytes, s);
+        else
+          hpk::check_hip(hipMemcpyAsync(peer_win, src, bytes,
+                                        hipMemcpyDeviceToDevice, s),
+                         "ipc put");
+        hpk::check_hip(hipStreamSynchronize(s), "put sync");
+      }
+      sh->dt[rank] = now_s() - t0;
+      if (!sh->bar.wait()) return 3; // fence: epoch close (puts visible)
+      // clock union: global interval = MAX over ranks (all started together
+      // at the opening fence — reference MPI_Reduce MIN/MAX, :49-51)
+      double g_dt = 0.0;
+      for (int r = 0; r < size; ++r) g_dt = std::max(g_dt, sh->dt[r]);
+      best = std::min(best, g_dt);
+      if (!sh->bar.wait()) return 3; // dt array consumed before next write
     }
-    std::printf("ipc Unidirectional Bandwidth: %.2f GB/s (one-sided put, "
-                "%.1f MB, min over %d iters, dev%d->dev%d)\n",
-                bytes / 1e9 / best, bytes / 1e6, kIters, dev, 0);
-    hpk::ipc_close_handle(target);
-    std::exit(0);
+    // verify: my window must hold my pair-peer's payload (odd ranks in the
+    // uni phase, every paired rank in the bidi phase)
+    bool target = paired && ((rank % 2 == 1) || bidir);
+    if (target) verify(win, n, sh->src_sum[peer], "ipc put");
+    if (rank == 0) {
+      int npairs = size / 2;
+      double gb = (double)bytes * npairs * (bidir ? 2 : 1) / 1e9;
+      std::printf("ipc %s Bandwidth: %.2f GB/s (pairs=%d, %.1f MB each, "
+                  "min over %d iters, one-sided put, fence epochs)\n",
+                  bidir ? "Bidirectional" : "Unidirectional", gb / best,
+                  npairs, bytes / 1e6, kIters);
+    }
+    if (!sh->bar.wait()) return 3; // phase boundary
   }
+  if (paired) hpk::ipc_close_handle(peer_win);
+  (void)hipFree(win);
+  (void)hipFree(src);
+  return rc;
+}
 
-  // parent = rank 0 (owner of the exposed window)
-  close(sv[1]);
-  hpk::check_hip(hipSetDevice(0), "parent set dev");
-  float* win = nullptr;
-  hpk::check_hip(hipMalloc(&win, bytes), "parent win");
-  auto hv = hpk::ipc_get_handle(win);
-  if (write(sv[0], hv.data(), hv.size()) != (ssize_t)hv.size()) return 1;
-  double expected = 0.0;
-  if (read(sv[0], &expected, sizeof(expected)) != sizeof(expected)) return 1;
-
-  for (int it = 0; it < kIters; ++it) {
-    char token = 'g';
-    if (write(sv[0], &token, 1) != 1) return 1;
-    double dt;
-    if (read(sv[0], &dt, sizeof(dt)) != sizeof(dt)) return 1;
+int run_ipc(const char* self, size_t n, int nranks) {
+  // parent stays HIP-free (launch_util.h invariant); fork BEFORE any HIP
+  // call — an initialized HIP runtime does not survive fork().
+  int ndev = hpk_launch::probe_device_count(self);
+  if (ndev == 0) {
+    std::fprintf(stderr, "no HIP devices\n");
+    return 1;
   }
-  verify(win, n, expected, "ipc put");
-  int st = 0;
-  waitpid(pid, &st, 0);
-  std::printf("# ipc window verified on owner side\n");
-  return WIFEXITED(st) ? WEXITSTATUS(st) : 1;
+  // default: one rank per GPU; 1-GPU boxes oversubscribe to 2 ranks so the
+  // cross-process protocol still runs end to end
+  int size = nranks > 0 ? nranks : (ndev >= 2 ? ndev : 2);
+  size -= size % 2; // pairs
+  size = std::max(2, std::min(size, kMaxIpcRanks));
+
+  IpcShared* sh = hpk_launch::map_shared_struct<IpcShared>();
+  if (!sh) {
+    std::fprintf(stderr, "shared map failed\n");
+    return 1;
+  }
+  sh->bar.size = size;
+  std::printf("# ipc engine: %d ranks on %d GPU(s) (%s)\n", size, ndev,
+              size > ndev ? "oversubscribed" : "one rank per GPU");
+  return hpk_launch::fork_workers(
+      size, [&](int rank) { return ipc_rma_worker(rank, size, ndev, sh, n); });
 }
 
 // ---------------------------------------------------------------------------
@@ -367,6 +413,7 @@ int run_rccl(const char* self, size_t n) {
 int main(int argc, char* argv[]) {
   std::string engine = "peer";
   size_t n = kDefaultN;
+  int nranks = -1;
   for (int i = 1; i < argc; ++i) {
     std::string s = argv[i];
     auto next = [&]() -> const char* {
@@ -378,13 +425,15 @@ int main(int argc, char* argv[]) {
       return 0;
     } else if (s == "--engine") engine = next();
     else if (s == "-n" || s == "--floats") n = std::strtoull(next(), nullptr, 10);
+    else if (s == "--ranks") nranks = std::atoi(next());
     else {
-      std::printf("Usage: %s [--engine peer|ipc|rccl] [--floats N]\n", argv[0]);
+      std::printf("Usage: %s [--engine peer|ipc|rccl] [--floats N] "
+                  "[--ranks R]\n", argv[0]);
       return (s == "-h" || s == "--help") ? 0 : 1;
     }
   }
   if (engine == "peer") return run_peer(n, true);
-  if (engine == "ipc") return run_ipc(n);
+  if (engine == "ipc") return run_ipc(argv[0], n, nranks);
   if (engine == "rccl") return run_rccl(argv[0], n);
   std::fprintf(stderr, "unknown engine '%s'\n", engine.c_str());
   return 1;

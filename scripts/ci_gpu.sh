@@ -27,9 +27,17 @@ echo "== miniapps"
 ./bin/hpk_allreduce -p 22 -i 2 --algo ring || rc=1
 ./bin/hpk_allreduce -p 22 -i 2 --algo pipeline || rc=1
 ./bin/hpk_allreduce -p 22 -i 2 --algo rccl -t int || rc=1
+# oversubscribed multi-rank exchange paths (run the REAL ring / all-pairs
+# RMA protocol even on a 1-GPU lease)
+./bin/hpk_allreduce -p 20 -i 2 --transport ipc -n 4 || rc=1
 ./bin/hpk_p2p --engine ipc --floats $((1 << 22)) || rc=1
+./bin/hpk_p2p --engine ipc --floats $((1 << 22)) --ranks 4 || rc=1
 ./bin/hpk_interop || rc=1
 ./bin/hpk_membench --quick || rc=1
+# graph_explicit mode + per-command device times in graph mode (r2)
+./bin/hpk_conc graph_explicit --repetitions 5 --enable_profiling \
+    --globalsize_default_memory $((1 << 24)) \
+    --commands C D2D || rc=1
 
 echo "== cmake + ctest harness (the reference's build contract)"
 rm -rf build-cmake && mkdir -p build-cmake
