@@ -28,6 +28,21 @@ LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
 EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)
 BINARIES   := $(BIN)/hpk_conc $(BIN)/hpk_topology $(BIN)/hpk_allreduce $(BIN)/hpk_p2p $(BIN)/hpk_interop $(BIN)/hpk_membench
 
+# Real-MPI twins (MPICH 3.3.2 ships in /opt/conda — not GPU-aware, so the
+# miniapps use pinned-direct / staged-device buffer modes; see PARITY.md).
+# rpath order matters: the system libstdc++ must shadow conda's old one.
+MPI_HOME   ?= /opt/conda
+HAVE_MPI   := $(wildcard $(MPI_HOME)/include/mpi.h)
+MPI_CFLAGS := -I$(MPI_HOME)/include
+# libmpi by absolute path (NOT -L$(MPI_HOME)/lib: that would let the linker
+# resolve libstdc++ from conda's old copy); rpath keeps system dirs first
+# so the same shadowing cannot happen at runtime.
+MPI_LD     := $(MPI_HOME)/lib/libmpi.so -Wl,-rpath,/usr/lib/x86_64-linux-gnu \
+              -Wl,-rpath,/opt/rocm/lib -Wl,-rpath,$(MPI_HOME)/lib
+ifneq ($(HAVE_MPI),)
+BINARIES   += $(BIN)/hpk_mpi_allreduce $(BIN)/hpk_mpi_p2p
+endif
+
 .PHONY: all ext bins clean
 all: ext bins
 ext: $(EXT_SO)
@@ -62,6 +77,14 @@ $(BIN)/hpk_interop: $(BUILD)/interop_main.o $(LIB_OBJS) | $(BIN)
 
 $(BIN)/hpk_membench: $(BUILD)/membench_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS)
+
+$(BUILD)/mpi_allreduce_main.o $(BUILD)/mpi_p2p_main.o: CXXFLAGS += $(MPI_CFLAGS)
+
+$(BIN)/hpk_mpi_allreduce: $(BUILD)/mpi_allreduce_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) $(MPI_LD)
+
+$(BIN)/hpk_mpi_p2p: $(BUILD)/mpi_p2p_main.o $(LIB_OBJS) | $(BIN)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) $(MPI_LD)
 
 $(BUILD):
 	mkdir -p $(BUILD)

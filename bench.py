@@ -29,8 +29,14 @@ import time
 def _component(components: dict, key: str, fn):
     """Run one diagnostic component; a failure degrades to a null field and
     a stderr note instead of killing the whole bench JSON line (the 8-GPU
-    scale run must survive any single component failing — VERDICT r1 #1)."""
+    scale run must survive any single component failing — VERDICT r1 #1).
+
+    HPK_BENCH_FAULT=<key> injects a failure into that component (CI
+    rehearsal of the degradation path; symmetric across ranks, so no rank
+    is stranded in a half-entered collective)."""
     try:
+        if os.environ.get("HPK_BENCH_FAULT") == key:
+            raise RuntimeError("injected fault (HPK_BENCH_FAULT)")
         val = fn()
     except Exception as e:  # noqa: BLE001 — any component error is non-fatal
         print(f"# component {key} failed: {type(e).__name__}: {e}",

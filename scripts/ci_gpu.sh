@@ -39,6 +39,18 @@ echo "== miniapps"
     --globalsize_default_memory $((1 << 24)) \
     --commands C D2D || rc=1
 
+echo "== real-MPI miniapps (MPICH, pinned-direct + staged-device modes)"
+if [ -x /opt/conda/bin/mpirun ] && [ -x bin/hpk_mpi_allreduce ]; then
+  /opt/conda/bin/mpirun -np 4 ./bin/hpk_mpi_allreduce -H -p 20 -i 2 || rc=1
+  /opt/conda/bin/mpirun -np 4 ./bin/hpk_mpi_allreduce -D -a -p 20 -i 2 || rc=1
+  /opt/conda/bin/mpirun -np 4 ./bin/hpk_mpi_p2p --engine isend -H \
+      --floats $((1 << 22)) || rc=1
+  /opt/conda/bin/mpirun -np 4 ./bin/hpk_mpi_p2p --engine win -H \
+      --floats $((1 << 22)) || rc=1
+else
+  echo "# skipped (no mpirun or binary)"
+fi
+
 echo "== cmake + ctest harness (the reference's build contract)"
 rm -rf build-cmake && mkdir -p build-cmake
 (cd build-cmake && CXX=/opt/rocm/bin/hipcc cmake .. > cmake.log 2>&1 \

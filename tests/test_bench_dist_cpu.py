@@ -67,6 +67,8 @@ def test_p2p_latency_sweep_cli(dist_env):
 
 
 def test_allreduce_sweep_cli(dist_env):
+    import os
+
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node", "2",
@@ -76,12 +78,17 @@ def test_allreduce_sweep_cli(dist_env):
         "--min-mb", "0.01", "--max-mb", "0.02", "--iters", "2",
         "--algos", "rccl,ring,rsag",
     ]
+    # the channel setting must land in the CSV's channels column (the
+    # channel-study contract, scripts/run_allreduce_sweep.sh --channels)
+    env = dict(os.environ)
+    env["NCCL_MIN_NCHANNELS"] = "4"
+    env["NCCL_MAX_NCHANNELS"] = "4"
     res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
-                         cwd=REPO)
+                         cwd=REPO, env=env)
     assert res.returncode == 0, res.stdout + res.stderr
-    assert "algo,bytes,time_s,alg_GBps,bus_GBps" in res.stdout
+    assert "algo,channels,bytes,time_s,alg_GBps,bus_GBps" in res.stdout
     for algo in ("rccl", "ring", "rsag"):
-        assert f"\n{algo}," in res.stdout or res.stdout.startswith(f"{algo},"), algo
+        assert f"\n{algo},4," in res.stdout, algo
 
 
 def test_bench_torchrun_cpu_world8(dist_env):
@@ -102,3 +109,41 @@ def test_bench_torchrun_cpu_world8(dist_env):
     assert data["n_gpus"] == 8
     assert data["config"]["parallelism"] == "dp8"
     assert data["components"]["p2p_checksum_ok"] is True
+
+
+def test_bench_component_fault_degrades_to_null(dist_env):
+    """Fault-injection rehearsal: a failing diagnostic
+    component must yield a null JSON field and rc 0, not kill the bench."""
+    import os
+
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        str(REPO / "bench.py"), "--gpus", "2", "--steps", "2",
+        "--warmup", "1", "--cpu",
+    ]
+    env = dict(os.environ)
+    env["HPK_BENCH_FAULT"] = "pingpong_us"
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO, env=env)
+    assert res.returncode == 0, res.stdout + res.stderr
+    data = json.loads([l for l in res.stdout.splitlines()
+                       if l.startswith("{")][0])
+    assert data["components"]["pingpong_us"] is None
+    # the OTHER components still measured
+    assert data["components"]["p2p_checksum_ok"] is True
+    assert "injected fault" in res.stderr + res.stdout
+
+
+def test_bench_policy_flag_cpu():
+    """--policy must parse and land in the config record."""
+    res = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "2",
+         "--warmup", "1", "--cpu", "--policy", "spread"],
+        capture_output=True, text=True, timeout=120, cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    data = json.loads([l for l in res.stdout.splitlines()
+                       if l.startswith("{")][-1])
+    assert data["config"]["placement_policy"] == "spread"

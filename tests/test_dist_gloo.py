@@ -180,3 +180,25 @@ def body_sweep(rank, world):
 
 def test_sweep_bench_algo(dist_env):
     run_dist(2, "body_sweep", int(dist_env["MASTER_PORT"]))
+
+
+def body_policy_sweep_row(rank, world):
+    """The policy-sweep measurement path on gloo: the CSV
+    row must carry the policy, the world size, and one device per rank."""
+    from hpc_patterns_amd.parallel.policy_sweep import measure_policy_row
+
+    row = measure_policy_row("compact", 4096 * 4, iters=2,
+                             device=torch.device("cpu"))
+    parts = row.split(",")
+    assert parts[0] == "compact" and parts[1] == str(world)
+    assert parts[2] == "gloo"
+    assert len(parts[7].split("+")) == world
+    return True
+
+
+def test_policy_sweep_row(dist_env):
+    run_dist(2, "body_policy_sweep_row", int(dist_env["MASTER_PORT"]))
+
+
+def test_policy_sweep_row_world4(dist_env):
+    run_dist(4, "body_policy_sweep_row", int(dist_env["MASTER_PORT"]))
