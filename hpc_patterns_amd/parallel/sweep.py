@@ -96,8 +96,9 @@ def main(argv=None) -> int:
         nbytes = int(args.min_mb * 1e6)
         while nbytes <= max_bytes:
             largest = nbytes * 2 > max_bytes
+            sampling = args.xgmi_sample and largest and rank == 0
             xgmi_before = None
-            if args.xgmi_sample and largest and rank == 0:
+            if sampling:
                 from ..utils import xgmi
 
                 xgmi_before = xgmi.sample_xgmi()
@@ -107,7 +108,7 @@ def main(argv=None) -> int:
                 bus_bw = 2 * (world - 1) / world * alg_bw if world > 1 else alg_bw
                 print(f"{algo},{channels},{nbytes},{t:.6f},{alg_bw:.2f},"
                       f"{bus_bw:.2f}", flush=True)
-                if xgmi_before is not None:
+                if sampling:
                     from ..utils import xgmi
 
                     d = xgmi.delta(xgmi_before, xgmi.sample_xgmi())
@@ -117,7 +118,10 @@ def main(argv=None) -> int:
                               + " ".join(f"{k}={v:.0f}" for k, v in top),
                               flush=True)
                     else:
-                        print("# xgmi_delta unavailable", flush=True)
+                        # 1-GPU boxes report read/write as N/A; the links
+                        # only count on a multi-GPU node
+                        print("# xgmi_delta unavailable (no numeric link "
+                              "counters on this node)", flush=True)
             nbytes *= 2
     dist.destroy_process_group()
     return 0
