@@ -28,7 +28,7 @@ do
     export $envs
     for engine in auto sdma; do
     echo "export $envs HPK_COPY_ENGINE=$engine"
-    for mode in in_order graph host_threads; do
+    for mode in in_order graph graph_explicit host_threads; do
         args=""
         for c in "${LCOMMANDS[@]}"; do args+=" --commands $c"; done
         $BIN "$mode" --copy_engine "$engine" --repetitions 5 $args
