@@ -147,3 +147,25 @@ def test_bench_policy_flag_cpu():
     data = json.loads([l for l in res.stdout.splitlines()
                        if l.startswith("{")][-1])
     assert data["config"]["placement_policy"] == "spread"
+
+
+def test_policy_sweep_cli_writes_csv(dist_env, tmp_path):
+    """policy_sweep CLI end to end on gloo: CSV file with header + one row."""
+    import os
+
+    csv = tmp_path / "policy.csv"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1",
+        "--master-port", dist_env["MASTER_PORT"],
+        "-m", "hpc_patterns_amd.parallel.policy_sweep",
+        "--policy", "spread", "--floats", "4096", "--iters", "2",
+        "--backend", "gloo", "--csv", str(csv),
+    ]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=240,
+                         cwd=REPO)
+    assert res.returncode == 0, res.stdout + res.stderr
+    lines = csv.read_text().strip().splitlines()
+    assert lines[0].startswith("policy,world,engine,bytes")
+    assert lines[1].startswith("spread,2,gloo,16384,")
