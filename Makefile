@@ -78,7 +78,11 @@ $(BIN)/hpk_interop: $(BUILD)/interop_main.o $(LIB_OBJS) | $(BIN)
 $(BIN)/hpk_membench: $(BUILD)/membench_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS)
 
-$(BUILD)/mpi_allreduce_main.o $(BUILD)/mpi_p2p_main.o: CXXFLAGS += $(MPI_CFLAGS)
+# dedicated rule (NOT a target-specific CXXFLAGS append: `make asan`
+# overrides CXXFLAGS on the command line, which would drop the append) —
+# the shorter-stem pattern beats the generic %_main.o rule
+$(BUILD)/mpi_%_main.o: cpp/mpi_%_main.cpp $(NATIVE)/include/hpk.h | $(BUILD)
+	$(HIPCC) $(CXXFLAGS) $(MPI_CFLAGS) -c $< -o $@
 
 $(BIN)/hpk_mpi_allreduce: $(BUILD)/mpi_allreduce_main.o $(LIB_OBJS) | $(BIN)
 	$(HIPCC) --offload-arch=$(GPU_ARCH) $^ -o $@ $(LDFLAGS) $(MPI_LD)
