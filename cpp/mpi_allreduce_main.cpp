@@ -289,6 +289,13 @@ int main(int argc, char* argv[]) {
       return (s == "-h" || s == "--help") ? 0 : 1;
     }
   }
+  if (cfg.p > 30) {
+    // MPI counts are int; 2^31 elements would overflow the cast
+    if (rank == 0)
+      std::fprintf(stderr, "-p must be <= 30 (MPI int count limit)\n");
+    MPI_Finalize();
+    return 1;
+  }
   // reference guard (allreduce-mpi-sycl.cpp:95-97) relaxed to even >= 2 so
   // a 2-rank CPU smoke remains possible; the ctest registration uses -np 4.
   if (size % 2 != 0 || size < 2) {

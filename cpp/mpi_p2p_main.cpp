@@ -30,6 +30,7 @@
 #include <algorithm>
 #include <chrono>
 #include <cmath>
+#include <cstdint>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -208,6 +209,12 @@ int main(int argc, char* argv[]) {
   }
   if (engine != "isend" && engine != "win") {
     if (rank == 0) std::fprintf(stderr, "unknown engine '%s'\n", engine.c_str());
+    MPI_Finalize();
+    return 1;
+  }
+  if (n > (size_t)INT32_MAX) {
+    if (rank == 0)
+      std::fprintf(stderr, "--floats must be <= 2^31-1 (MPI int count)\n");
     MPI_Finalize();
     return 1;
   }

@@ -38,6 +38,14 @@ KNOBS = [
     Knob("HPK_PINNED_FLAGS", "pinned-alloc flavour for H buffers: default|nc|wc", "suite"),
     Knob("HPK_LAUNCH_TIMEOUT", "fork-launcher watchdog seconds (default 600)", "suite"),
     Knob("HPK_NGPUS", "GPU-count override for gpu_mapping.sh", "suite"),
+    Knob("HPK_COPY_STREAM_PRIORITY", "high = raise copy-pool stream priority "
+         "(co-scheduling-limited pods)", "suite"),
+    Knob("HPK_PLACEMENT_POLICY", "default rank->GPU policy for bench.py: "
+         "compact|spread|topo", "suite"),
+    Knob("HPK_NCCL_TIMEOUT_S", "process-group collective timeout for bench.py "
+         "(default 300)", "suite"),
+    Knob("HPK_BENCH_FAULT", "fault injection: named bench component raises "
+         "(CI rehearsal of null-field degradation)", "suite"),
     Knob("MASTER_ADDR", "torch.distributed rendezvous address (use 127.0.0.1)", "suite"),
     Knob("MASTER_PORT", "torch.distributed rendezvous port", "suite"),
 ]
