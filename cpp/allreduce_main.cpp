@@ -187,6 +187,8 @@ struct IpcRing {
   }
 
   void sendrecv(T* send, T* recv, size_t n, hipStream_t stream) {
+    if (!ok) return; // a fence already timed out (lost sibling) — stop
+                     // issuing puts; the worker reports the failure
     // my recv allocation index == right's recv allocation index (lockstep)
     int idx = (recv == alloc1) ? 1 : 0;
     void* dst = right_buf[idx];

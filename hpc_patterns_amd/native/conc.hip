@@ -201,6 +201,18 @@ ConcResult conc_bench(const std::string& mode,
 
   // ---- build commands + buffers ----
   std::vector<Command> cmds(ncmds);
+  // GiB-scale buffers must not leak into a long-lived python process when
+  // the engine throws (e.g. a capture error): the catch at the bottom
+  // frees them after a device sync. (Events/graphs are byte-scale and the
+  // capture paths abort-guard themselves.)
+  auto free_buffers = [&]() {
+    for (auto& c : cmds) {
+      c.src.free();
+      c.dst.free();
+      c.out.free();
+    }
+  };
+  try {
   for (int i = 0; i < ncmds; ++i) {
     Command& c = cmds[i];
     c.name = commands[i];
@@ -536,12 +548,13 @@ ConcResult conc_bench(const std::string& mode,
     }
   }
   // streams belong to the process-lifetime pool — not destroyed here
-  for (auto& c : cmds) {
-    const_cast<Buffer&>(c.src).free();
-    const_cast<Buffer&>(c.dst).free();
-    const_cast<Buffer&>(c.out).free();
-  }
+  free_buffers();
   return res;
+  } catch (...) {
+    (void)hipDeviceSynchronize();
+    free_buffers();
+    throw;
+  }
 }
 
 } // namespace hpk
