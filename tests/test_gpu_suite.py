@@ -283,3 +283,37 @@ def test_bin_mpi_p2p_pipelined_staged():
     assert res.returncode == 0, res.stdout + res.stderr
     assert "mpi-isend-pipe Unidirectional Bandwidth" in res.stdout
     assert "mpi-isend-pipe Bidirectional Bandwidth" in res.stdout
+
+
+def test_pipelined_ring_stream_choreography_loopback(monkeypatch):
+    """The two-stream pipelined ring (comm ops + per-chunk accumulate on a
+    dedicated compute stream) is otherwise unreachable before a multi-GPU
+    run: NCCL refuses two ranks on one device. This swaps the transport
+    for an in-process loopback (recv := send on the current stream) while
+    keeping every stream/wait/accumulate line of the real path; wrong
+    cross-stream ordering would corrupt the deterministic result."""
+    import torch.distributed as real_dist
+
+    from hpc_patterns_amd.parallel import ring
+
+    class _FakeReq:
+        def wait(self):
+            pass
+
+    def fake_batch(ops):
+        sends = [op.tensor for op in ops if op.op is real_dist.isend]
+        recvs = [op.tensor for op in ops if op.op is real_dist.irecv]
+        for s, r in zip(sends, recvs):
+            r.copy_(s)  # ordered on the calling stream, like the comm op
+        return [_FakeReq()]
+
+    # pretend world=2 with self as both neighbours
+    monkeypatch.setattr(ring, "_ring_neighbours", lambda group=None: (0, 2, 0, 0))
+    monkeypatch.setattr(ring.dist, "batch_isend_irecv", fake_batch)
+
+    t = torch.arange(1 << 20, dtype=torch.float32, device="cuda")
+    expect = t * 2  # one loopback exchange step: chunks += their own send
+    ring.ring_allreduce_pipelined(t.clone(), n_chunks=8)  # smoke alias path
+    out = ring.ring_allreduce_pipelined(t, n_chunks=8)
+    torch.cuda.synchronize()
+    assert torch.equal(out, expect)
