@@ -292,7 +292,7 @@ def test_pipelined_ring_stream_choreography_loopback(monkeypatch):
     for an in-process loopback (recv := send on the current stream) while
     keeping every stream/wait/accumulate line of the real path; wrong
     cross-stream ordering would corrupt the deterministic result."""
-    import torch.distributed as real_dist
+    from types import SimpleNamespace
 
     from hpc_patterns_amd.parallel import ring
 
@@ -301,15 +301,21 @@ def test_pipelined_ring_stream_choreography_loopback(monkeypatch):
             pass
 
     def fake_batch(ops):
-        sends = [op.tensor for op in ops if op.op is real_dist.isend]
-        recvs = [op.tensor for op in ops if op.op is real_dist.irecv]
+        sends = [op.tensor for op in ops if op.op == "isend"]
+        recvs = [op.tensor for op in ops if op.op == "irecv"]
         for s, r in zip(sends, recvs):
             r.copy_(s)  # ordered on the calling stream, like the comm op
         return [_FakeReq()]
 
-    # pretend world=2 with self as both neighbours
+    fake_dist = SimpleNamespace(
+        isend="isend", irecv="irecv",
+        P2POp=lambda op, tensor, peer, group=None: SimpleNamespace(
+            op=op, tensor=tensor),
+        batch_isend_irecv=fake_batch,
+    )
+    # pretend world=2 with self as both neighbours; no real process group
     monkeypatch.setattr(ring, "_ring_neighbours", lambda group=None: (0, 2, 0, 0))
-    monkeypatch.setattr(ring.dist, "batch_isend_irecv", fake_batch)
+    monkeypatch.setattr(ring, "dist", fake_dist)
 
     t = torch.arange(1 << 20, dtype=torch.float32, device="cuda")
     expect = t * 2  # one loopback exchange step: chunks += their own send
