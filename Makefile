@@ -22,7 +22,7 @@ CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(GPU_ARCH) -I$(NATIVE) -Wall
 LDEXTRA    ?=
 LDFLAGS    := $(LDEXTRA) -L/opt/rocm/lib -lrocm_smi64 -lrocprofiler-sdk-roctx -lhsa-runtime64
 
-LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip $(NATIVE)/trace.hip $(NATIVE)/sdma.hip $(NATIVE)/staged.hip
+LIB_SRCS   := $(NATIVE)/kernels.hip $(NATIVE)/gemm.hip $(NATIVE)/conc.hip $(NATIVE)/topo.hip $(NATIVE)/ipc.hip $(NATIVE)/trace.hip $(NATIVE)/sdma.hip $(NATIVE)/staged.hip
 LIB_OBJS   := $(patsubst $(NATIVE)/%.hip,$(BUILD)/%.o,$(LIB_SRCS))
 
 EXT_SO     := hpc_patterns_amd/_hpk$(EXT_SUFFIX)

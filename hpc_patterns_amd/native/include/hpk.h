@@ -64,6 +64,15 @@ void launch_acc_f32(float* dst, const float* src, size_t n, hipStream_t stream);
 void launch_acc_f32_nt(float* dst, const float* src, size_t n,
                        hipStream_t stream);
 
+// K7 (r2, beyond-parity showcase): LDS-tiled bf16 MFMA GEMM —
+// C[M,N] (fp32) = A[M,K] x B[N,K]^T, both operands bf16 K-contiguous.
+// 128x128 tile / 4 waves / v_mfma_f32_16x16x32_bf16, one-buffer
+// global_load_lds staging, optional bijective XCD workgroup swizzle.
+// Requires M,N % 128 == 0 and K % 64 == 0 (throws otherwise).
+void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
+                         long N, long K, hipStream_t stream,
+                         int xcd_swizzle = 1);
+
 // Exact double-precision sum of n floats. Synchronizes `stream`.
 // Replaces the reference's O(N log N) host sort+sum checksum
 // (peer2pear.cpp:56-63) with an order-independent exact device reduction.

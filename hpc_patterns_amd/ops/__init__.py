@@ -105,3 +105,27 @@ def iota_checksum(n: int) -> float:
 
     # float32 rounding of the iota values, summed exactly in float64
     return float(np.arange(n, dtype=np.float32).astype(np.float64).sum())
+
+
+def gemm_bf16(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
+              stream=None, xcd_swizzle: bool = True) -> None:
+    """K7 (r2): C[M,N] fp32 = A[M,K] @ B[N,K]^T, A/B bf16 K-contiguous.
+
+    Hand-written LDS-tiled v_mfma_f32_16x16x32_bf16 kernel (native/gemm.hip):
+    128x128 tile per 4-wave workgroup, 16-byte global_load_lds staging,
+    bijective XCD workgroup swizzle. Requires M,N multiples of 128 and K a
+    multiple of 64.
+    """
+    if c.dtype != torch.float32 or a.dtype != torch.bfloat16 \
+            or b.dtype != torch.bfloat16:
+        raise TypeError("c must be fp32; a, b must be bf16")
+    for t, name in ((c, "c"), (a, "a"), (b, "b")):
+        if not t.is_cuda or not t.is_contiguous() or t.dim() != 2:
+            raise TypeError(f"{name} must be a contiguous 2-D CUDA tensor")
+    m, k = a.shape
+    n, kb = b.shape
+    if kb != k or c.shape != (m, n):
+        raise ValueError(f"shape mismatch: A{tuple(a.shape)} B{tuple(b.shape)}"
+                         f" C{tuple(c.shape)}")
+    native().gemm_bf16_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                          m, n, k, _stream_handle(stream), int(xcd_swizzle))

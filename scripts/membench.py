@@ -98,6 +98,22 @@ def main():
     flops = 2048 * 20000 * 16384
     print(f"mfma        {t*1e3:9.3f} ms  {flops/t/1e12:9.2f} TFLOP/s bf16")
 
+    print("\n# K7 LDS-tiled bf16 GEMM (C = A x B^T, fp32 accumulate, "
+          "random [-1,1) operands)")
+    from hpc_patterns_amd import ops
+
+    for sz in (2048, 4096, 8192):
+        a = (torch.rand(sz, sz, device=dev) * 2 - 1).to(torch.bfloat16)
+        b = (torch.rand(sz, sz, device=dev) * 2 - 1).to(torch.bfloat16)
+        c = torch.empty(sz, sz, dtype=torch.float32, device=dev)
+        for sw in (True, False):
+            t = time_gpu(lambda: ops.gemm_bf16(c, a, b, xcd_swizzle=sw))
+            fl = 2.0 * sz * sz * sz
+            tag = "swz" if sw else "lin"
+            print(f"gemm {sz:5d} {tag} {t*1e3:9.3f} ms  "
+                  f"{fl/t/1e12:9.1f} TFLOP/s bf16")
+        del a, b, c
+
 
 if __name__ == "__main__":
     main()

@@ -295,3 +295,71 @@ def test_staged_copy_chunk_boundaries(ops, dev, nbytes):
     hpk.staged_copy(out.ctypes.data, dbuf, nbytes, 0, 0, False)
     assert np.array_equal(out, src), nbytes
     hpk.hip_free(dbuf)
+
+
+# ---------------------------------------------------------------------------
+# K7 (r2): LDS-tiled bf16 MFMA GEMM
+# ---------------------------------------------------------------------------
+
+def test_gemm_bf16_identity_asymmetric():
+    """A = I with an ASYMMETRIC B catches any row/col-swapped fragment or
+    epilogue mapping (the guide's A=I-check): C = I @ B^T = B^T exactly."""
+    from hpc_patterns_amd import ops
+
+    torch.manual_seed(7)
+    m = n = k = 128
+    a = torch.eye(m, k, device="cuda").to(torch.bfloat16)
+    b = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(c, a, b)
+    torch.cuda.synchronize()
+    assert torch.equal(c, b.t().float()), (c - b.t().float()).abs().max()
+
+
+def test_gemm_bf16_exact_integers():
+    """Small-integer payloads make every product and partial sum exactly
+    representable in fp32, so the MFMA result must EQUAL the torch fp32
+    reference bit for bit regardless of summation order."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(11)
+    m, n, k = 256, 384, 512
+    a = torch.randint(-4, 5, (m, k), generator=g).to(torch.bfloat16).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).to(torch.bfloat16).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(c, a, b)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+@pytest.mark.parametrize("swizzle", [True, False])
+def test_gemm_bf16_large_close(swizzle):
+    """1024^3 random normal: products of bf16 values are exact in fp32, so
+    the only divergence from the torch fp32 reference is summation order —
+    tight allclose."""
+    from hpc_patterns_amd import ops
+
+    torch.manual_seed(3)
+    m = n = k = 1024
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+    b = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(c, a, b, xcd_swizzle=swizzle)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.allclose(c, ref, rtol=1e-4, atol=1e-3), \
+        (c - ref).abs().max()
+
+
+def test_gemm_bf16_shape_guards():
+    from hpc_patterns_amd import ops
+
+    a = torch.zeros(128, 64, dtype=torch.bfloat16, device="cuda")
+    b = torch.zeros(128, 64, dtype=torch.bfloat16, device="cuda")
+    c = torch.zeros(128, 128, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(c, a, b)  # minimal legal shape
+    with pytest.raises((RuntimeError, ValueError)):
+        ops.gemm_bf16(c, a[:, :32].contiguous(), b[:, :32].contiguous())
+    with pytest.raises(TypeError):
+        ops.gemm_bf16(c, a.float(), b)
