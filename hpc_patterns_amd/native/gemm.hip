@@ -29,6 +29,7 @@
 
 #include <hip/hip_bf16.h>
 
+#include <cstdlib>
 #include <stdexcept>
 
 namespace hpk {
@@ -38,17 +39,47 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr int BM = 128, BN = 128, BK = 64;
-constexpr int WAVES_M = 2, WAVES_N = 2;       // 2x2 waves of 64x64 each
-constexpr int THREADS = WAVES_M * WAVES_N * 64;
-constexpr int MREP = 4, NREP = 4;             // 16x16 fragments per wave
 
 // one K-step's staging: A[128][64] + B[128][64] bf16 = 32 KiB
 constexpr int TILE_HALF = BM * BK;            // elements per operand tile
 
-__global__ __launch_bounds__(THREADS) void k_gemm_bf16_nt(
+// Cyclic-skew LDS layout: linear [128][64]-bf16 rows are 128 B, so the
+// 16 rows a ds_read_b128 lane group touches land on only two 4-bank
+// windows (8-way conflict; st_16x32 XOR halves it to 2-way — measured
+// SQ_LDS_BANK_CONFLICT still 0.5x of IDX_ACTIVE). Rotating each row's
+// 64-element K range by 8 elements per row PAIR puts all 16 rows of a
+// lane group on 16 DISTINCT 4-dword bank windows:
+//   LDS(row, k) = (row, (k + 8*((row>>1)&7)) & 63)
+// The 8-element granule keeps every 16-B glds chunk and every 8-element
+// fragment read contiguous, so the staging pre-applies the inverse on the
+// GLOBAL source chunk address while the LDS write stays lane-linear (the
+// global_load_lds requirement); the skew costs only slightly shuffled
+// global fetch order within each 128-B row.
+__device__ __forceinline__ long lds_skew(long e) { // tile elem -> LDS slot
+  long row = e >> 6, k = e & 63;
+  return (row << 6) | ((k + 8 * ((row >> 1) & 7)) & 63);
+}
+__device__ __forceinline__ long lds_unskew(long y) { // LDS slot -> tile elem
+  long row = y >> 6, k = y & 63;
+  return (row << 6) | ((k - 8 * ((row >> 1) & 7)) & 63);
+}
+
+// Wave-grid decomposition is a template knob: <2,2> = 4 waves of 64x64
+// (4x4 fragments, 190 VGPR+AGPR, 2 waves/SIMD), <2,4> = 8 waves of
+// 64x32 (4x2 fragments, 76 VGPR, 6 waves/SIMD). Measured on MI355X with
+// random operands: 8 waves wins everywhere (924 vs 765 TF at 8192^3) —
+// occupancy-driven latency hiding beats the bigger per-wave MFMA batch.
+// HPK_GEMM_WAVES=4|8 overrides for experiments.
+template <int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
     const __hip_bfloat16* __restrict__ B, int M, int N, int K,
     int tiles_n, int nwg, int xcd_swizzle) {
+  constexpr int THREADS = WAVES_M * WAVES_N * 64;
+  constexpr int MREP = BM / (WAVES_M * 16);
+  constexpr int NREP = BN / (WAVES_N * 16);
+  constexpr int WTM = BM / WAVES_M; // wave sub-tile rows
+  constexpr int WTN = BN / WAVES_N; // wave sub-tile cols
   __shared__ __hip_bfloat16 lds[2 * TILE_HALF]; // [A tile][B tile]
 
   int wg = (int)blockIdx.x;
@@ -74,14 +105,18 @@ __global__ __launch_bounds__(THREADS) void k_gemm_bf16_nt(
   // writes to (wave-uniform LDS base) + lane*16, so the LDS pointer we
   // pass is the WAVE chunk base and only the GLOBAL address is per-lane;
   // the row-major [128][64] tile image is lane-linear by construction.
-  const long elems_per_issue = (long)THREADS * 8;
+  constexpr long elems_per_issue = (long)THREADS * 8;
   f32x4 acc[MREP][NREP] = {};
 
+  constexpr int ISSUES = TILE_HALF / (THREADS * 8);
   for (int k0 = 0; k0 < K; k0 += BK) {
     __syncthreads(); // previous K-step's reads done before overwrite
-    for (int issue = 0; issue < 4; ++issue) {
+    for (int issue = 0; issue < ISSUES; ++issue) {
       long o_base = (long)issue * elems_per_issue + (long)wid * (64 * 8);
-      long o = o_base + (long)lane * 8; // this lane's element offset
+      (void)0;
+      // lane's LDS slot is o_base + lane*8 (lane-linear); fetch the global
+      // chunk that belongs at that slot under the skewed image
+      long o = lds_unskew(o_base + (long)lane * 8);
       int row = (int)(o / BK);
       int kk = (int)(o % BK);
       const __hip_bfloat16* ga = A + (brow + row) * (long)K + k0 + kk;
@@ -103,12 +138,12 @@ __global__ __launch_bounds__(THREADS) void k_gemm_bf16_nt(
       const int kfrag = kk + 8 * (lane >> 4);
       bf16x8 afrag[MREP], bfrag[NREP];
       for (int m = 0; m < MREP; ++m) {
-        int row = wr * 64 + m * 16 + (lane & 15);
-        afrag[m] = *(const bf16x8*)(la + row * BK + kfrag);
+        int row = wr * WTM + m * 16 + (lane & 15);
+        afrag[m] = *(const bf16x8*)(la + lds_skew(row * BK + kfrag));
       }
       for (int n = 0; n < NREP; ++n) {
-        int col = wc * 64 + n * 16 + (lane & 15);
-        bfrag[n] = *(const bf16x8*)(lb + col * BK + kfrag);
+        int col = wc * WTN + n * 16 + (lane & 15);
+        bfrag[n] = *(const bf16x8*)(lb + lds_skew(col * BK + kfrag));
       }
       for (int m = 0; m < MREP; ++m)
         for (int n = 0; n < NREP; ++n)
@@ -120,8 +155,8 @@ __global__ __launch_bounds__(THREADS) void k_gemm_bf16_nt(
   // epilogue: C/D mapping row = 4*(lane>>4)+r, col = lane&15
   for (int m = 0; m < MREP; ++m)
     for (int n = 0; n < NREP; ++n) {
-      long row0 = brow + wr * 64 + m * 16 + 4 * (lane >> 4);
-      long col = bcol + wc * 64 + n * 16 + (lane & 15);
+      long row0 = brow + wr * WTM + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * WTN + n * 16 + (lane & 15);
       for (int r = 0; r < 4; ++r)
         C[(row0 + r) * (long)N + col] = acc[m][n][r];
     }
@@ -137,9 +172,23 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
         "gemm_bf16_nt requires M,N % 128 == 0 and K % 64 == 0");
   int tiles_m = (int)(M / BM), tiles_n = (int)(N / BN);
   int nwg = tiles_m * tiles_n;
-  hipLaunchKernelGGL(k_gemm_bf16_nt, dim3(nwg), dim3(THREADS), 0, stream,
-                     C, (const __hip_bfloat16*)A, (const __hip_bfloat16*)B,
-                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+  // measured (profiles/gemm_r2 logs, random [-1,1) operands): 8 waves
+  // 777/924 TF at 4096^3/8192^3 vs 4 waves 686/765 — the 64x32 sub-tile
+  // variant's 76-VGPR budget lifts occupancy 2 -> 6 waves/SIMD and wins
+  // everywhere; it is the default.
+  int waves = 8;
+  if (const char* env = std::getenv("HPK_GEMM_WAVES")) waves = std::atoi(env);
+  if (waves == 8) {
+    hipLaunchKernelGGL((k_gemm_bf16_nt<2, 4>), dim3(nwg), dim3(512), 0,
+                       stream, C, (const __hip_bfloat16*)A,
+                       (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
+                       tiles_n, nwg, xcd_swizzle);
+  } else {
+    hipLaunchKernelGGL((k_gemm_bf16_nt<2, 2>), dim3(nwg), dim3(256), 0,
+                       stream, C, (const __hip_bfloat16*)A,
+                       (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
+                       tiles_n, nwg, xcd_swizzle);
+  }
   check_hip(hipGetLastError(), "launch_gemm_bf16_nt");
 }
 
