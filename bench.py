@@ -166,6 +166,29 @@ def _measure_d2d_copy_tbps(step, iters: int = 5) -> float:
     return nbytes / best / 1e12
 
 
+def _measure_gemm_tf(step, size: int = 4096, iters: int = 3) -> float:
+    """K7 LDS-tiled bf16 MFMA GEMM TFLOP/s at size^3 on random operands
+    (the r2 showcase kernel; 2*size^3 FLOP per call)."""
+    import time
+
+    import torch
+
+    from hpc_patterns_amd import ops
+
+    a = (torch.rand(size, size, device=step.device) * 2 - 1).to(torch.bfloat16)
+    b = (torch.rand(size, size, device=step.device) * 2 - 1).to(torch.bfloat16)
+    c = torch.empty(size, size, dtype=torch.float32, device=step.device)
+    ops.gemm_bf16(c, a, b)  # warmup
+    torch.cuda.synchronize()
+    best = float("inf")
+    for _ in range(iters):
+        t0 = time.perf_counter()
+        ops.gemm_bf16(c, a, b)
+        torch.cuda.synchronize()
+        best = min(best, time.perf_counter() - t0)
+    return 2.0 * size ** 3 / best / 1e12
+
+
 def _measure_mfma_tf(step, n_waves: int = 2048, tripcount: int = 20000,
                      iters: int = 3) -> float:
     """bf16 MFMA busy-loop TFLOP/s (v_mfma_f32_16x16x32_bf16 chains):
@@ -373,6 +396,8 @@ def main() -> int:
                    lambda: round(_measure_d2d_copy_tbps(step), 2))
         _component(components, "mfma_TFbf16",
                    lambda: round(_measure_mfma_tf(step), 1))
+        _component(components, "gemm_TFbf16",
+                   lambda: round(_measure_gemm_tf(step), 1))
     if world > 1:
         def _p2p():
             bw = pairwise_bandwidth(cfg["p2p_floats"] * 4, iters=5,
