@@ -30,6 +30,7 @@
 #include <hip/hip_bf16.h>
 
 #include <cstdlib>
+#include <string>
 #include <stdexcept>
 
 namespace hpk {
@@ -162,6 +163,109 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     }
 }
 
+// Double-buffered variant: the plain kernel's __syncthreads() after the
+// glds issues carries an implicit s_waitcnt vmcnt(0) that drains the DMA
+// queue before ANY wave crosses — the documented ~20% stall of the
+// two-barrier structure. Here K-tile t+1's DMA is issued BEFORE waiting
+// for tile t (FIFO per wave, so `s_waitcnt vmcnt(8)` retires exactly
+// tile t's 8 DMAs while t+1's stay in flight), and the barriers are raw
+// s_barrier + lgkmcnt(0) so nothing re-drains the queue. 64 KiB LDS.
+template <int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt_db(
+    float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
+    const __hip_bfloat16* __restrict__ B, int M, int N, int K,
+    int tiles_n, int nwg, int xcd_swizzle) {
+  constexpr int THREADS = WAVES_M * WAVES_N * 64;
+  constexpr int MREP = BM / (WAVES_M * 16);
+  constexpr int NREP = BN / (WAVES_N * 16);
+  constexpr int WTM = BM / WAVES_M;
+  constexpr int WTN = BN / WAVES_N;
+  __shared__ __hip_bfloat16 lds[2 * 2 * TILE_HALF]; // 2 buffers x (A|B)
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * BM;
+  const long bcol = (long)(wg % tiles_n) * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WAVES_N;
+  const int wc = wid % WAVES_N;
+
+  constexpr long elems_per_issue = (long)THREADS * 8;
+  constexpr int ISSUES = TILE_HALF / (THREADS * 8);
+  f32x4 acc[MREP][NREP] = {};
+
+  auto stage = [&](int buf, int k0) {
+    __hip_bfloat16* dst = lds + (long)buf * 2 * TILE_HALF;
+    for (int issue = 0; issue < ISSUES; ++issue) {
+      long o_base = (long)issue * elems_per_issue + (long)wid * (64 * 8);
+      long o = lds_unskew(o_base + (long)lane * 8);
+      int row = (int)(o / BK);
+      int kk = (int)(o % BK);
+      const __hip_bfloat16* ga = A + (brow + row) * (long)K + k0 + kk;
+      const __hip_bfloat16* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(dst + TILE_HALF + o_base),
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    const int cur = (k0 / BK) & 1;
+    const bool more = (k0 + BK) < K;
+    if (more) stage(cur ^ 1, k0 + BK); // next tile's DMA, other buffer
+    // retire exactly the CURRENT tile's DMAs (2*ISSUES per thread issued
+    // first; FIFO), leaving the prefetch in flight across the barrier
+    if (more)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * ISSUES) : "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier(); // current buffer complete for all waves
+
+    const __hip_bfloat16* la = lds + (long)cur * 2 * TILE_HALF;
+    const __hip_bfloat16* lb = la + TILE_HALF;
+    for (int kk = 0; kk < BK; kk += 32) {
+      const int kfrag = kk + 8 * (lane >> 4);
+      bf16x8 afrag[MREP], bfrag[NREP];
+      for (int m = 0; m < MREP; ++m) {
+        int row = wr * WTM + m * 16 + (lane & 15);
+        afrag[m] = *(const bf16x8*)(la + lds_skew(row * BK + kfrag));
+      }
+      for (int n = 0; n < NREP; ++n) {
+        int col = wc * WTN + n * 16 + (lane & 15);
+        bfrag[n] = *(const bf16x8*)(lb + lds_skew(col * BK + kfrag));
+      }
+      for (int m = 0; m < MREP; ++m)
+        for (int n = 0; n < NREP; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[m], bfrag[n], acc[m][n], 0, 0, 0);
+    }
+    // every wave done READING buf[cur] before the next iteration's
+    // prefetch overwrites it
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * WTM + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * WTN + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 } // namespace
 
 void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
@@ -178,7 +282,19 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
   // everywhere; it is the default.
   int waves = 8;
   if (const char* env = std::getenv("HPK_GEMM_WAVES")) waves = std::atoi(env);
-  if (waves == 8) {
+  const char* var = std::getenv("HPK_GEMM_VARIANT");
+  const bool dbuf = var && std::string(var) == "db";
+  if (dbuf && waves == 8) {
+    hipLaunchKernelGGL((k_gemm_bf16_nt_db<2, 4>), dim3(nwg), dim3(512), 0,
+                       stream, C, (const __hip_bfloat16*)A,
+                       (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
+                       tiles_n, nwg, xcd_swizzle);
+  } else if (dbuf) {
+    hipLaunchKernelGGL((k_gemm_bf16_nt_db<2, 2>), dim3(nwg), dim3(256), 0,
+                       stream, C, (const __hip_bfloat16*)A,
+                       (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
+                       tiles_n, nwg, xcd_swizzle);
+  } else if (waves == 8) {
     hipLaunchKernelGGL((k_gemm_bf16_nt<2, 4>), dim3(nwg), dim3(512), 0,
                        stream, C, (const __hip_bfloat16*)A,
                        (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
