@@ -266,7 +266,121 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt_db(
     }
 }
 
+// ---------------------------------------------------------------------------
+// K7-fp8: same GEMM at OCP fp8 e4m3 (__builtin_amdgcn_mfma_f32_16x16x32_
+// fp8_fp8 runs at the bf16 MFMA rate but the staging traffic halves).
+// A/B fragments are 8 packed fp8 in one i64 (2 VGPRs), read from LDS by
+// ds_read_b64. The [128][64]-byte rows give ds_read_b64 lane groups of 32
+// a 4-way bank conflict in linear layout; shifting each row's K range by
+// 16 elements per 4-row class — LDS(row,k) = (row, (k + 16*((row>>2)&3))
+// & 63) — plus the kfrag offset spreads all 32 lanes of a group over 32
+// distinct 2-dword windows (conflict-free).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ long lds_skew8(long e) { // tile elem -> LDS slot
+  long row = e >> 6, k = e & 63;
+  return (row << 6) | ((k + 16 * ((row >> 2) & 3)) & 63);
+}
+__device__ __forceinline__ long lds_unskew8(long y) {
+  long row = y >> 6, k = y & 63;
+  return (row << 6) | ((k - 16 * ((row >> 2) & 3)) & 63);
+}
+
+template <int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_fp8_nt(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, int M, int N, int K,
+    int tiles_n, int nwg, int xcd_swizzle) {
+  constexpr int THREADS = WAVES_M * WAVES_N * 64;
+  constexpr int MREP = BM / (WAVES_M * 16);
+  constexpr int NREP = BN / (WAVES_N * 16);
+  constexpr int WTM = BM / WAVES_M;
+  constexpr int WTN = BN / WAVES_N;
+  __shared__ unsigned char lds[2 * TILE_HALF]; // bytes: [A tile][B tile]
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * BM;
+  const long bcol = (long)(wg % tiles_n) * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WAVES_N;
+  const int wc = wid % WAVES_N;
+
+  constexpr long bytes_per_issue = (long)THREADS * 16;
+  constexpr int ISSUES = TILE_HALF / (THREADS * 16); // 1 byte/elem
+  static_assert(ISSUES >= 1, "tile too small for the staging plan");
+  f32x4 acc[MREP][NREP] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    for (int issue = 0; issue < ISSUES; ++issue) {
+      long o_base = (long)issue * bytes_per_issue + (long)wid * (64 * 16);
+      // one 16-B chunk = 16 fp8 elements spanning k-bits 0-3; the skew
+      // shift granule is 16 elements, so chunks stay contiguous
+      long o = lds_unskew8(o_base + (long)lane * 16);
+      int row = (int)(o / BK);
+      int kk = (int)(o % BK);
+      const unsigned char* ga = A + (brow + row) * (long)K + k0 + kk;
+      const unsigned char* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(lds + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(lds + TILE_HALF + o_base),
+          16, 0, 0);
+    }
+    __syncthreads();
+
+    const unsigned char* la = lds;
+    const unsigned char* lb = lds + TILE_HALF;
+    for (int kk = 0; kk < BK; kk += 32) {
+      const int kfrag = kk + 8 * (lane >> 4);
+      long afrag[MREP], bfrag[NREP];
+      for (int m = 0; m < MREP; ++m) {
+        int row = wr * WTM + m * 16 + (lane & 15);
+        afrag[m] = *(const long*)(la + lds_skew8(row * BK + kfrag));
+      }
+      for (int n = 0; n < NREP; ++n) {
+        int col = wc * WTN + n * 16 + (lane & 15);
+        bfrag[n] = *(const long*)(lb + lds_skew8(col * BK + kfrag));
+      }
+      for (int m = 0; m < MREP; ++m)
+        for (int n = 0; n < NREP; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              afrag[m], bfrag[n], acc[m][n], 0, 0, 0);
+    }
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * WTM + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * WTN + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 } // namespace
+
+void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
+                        long N, long K, hipStream_t stream,
+                        int xcd_swizzle) {
+  if (M % BM != 0 || N % BN != 0 || K % BK != 0)
+    throw std::runtime_error(
+        "gemm_fp8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  int tiles_m = (int)(M / BM), tiles_n = (int)(N / BN);
+  int nwg = tiles_m * tiles_n;
+  hipLaunchKernelGGL((k_gemm_fp8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,
+                     C, (const unsigned char*)A, (const unsigned char*)B,
+                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+  check_hip(hipGetLastError(), "launch_gemm_fp8_nt");
+}
 
 void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
                          long N, long K, hipStream_t stream,

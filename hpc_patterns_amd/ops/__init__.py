@@ -132,3 +132,26 @@ def gemm_bf16(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
                          f" C{tuple(c.shape)}")
     native().gemm_bf16_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
                           m, n, k, _stream_handle(stream), int(xcd_swizzle))
+
+
+def gemm_fp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
+             stream=None, xcd_swizzle: bool = False) -> None:
+    """K7-fp8: C[M,N] fp32 = A[M,K] @ B[N,K]^T, A/B torch.float8_e4m3fn.
+
+    The mfma_f32_16x16x32_fp8_fp8 twin of gemm_bf16 (same MFMA rate, half
+    the staging bytes; gfx950 fp8 is OCP e4m3 — exactly torch's
+    float8_e4m3fn). Same shape constraints as gemm_bf16.
+    """
+    if c.dtype != torch.float32 or a.dtype != torch.float8_e4m3fn \
+            or b.dtype != torch.float8_e4m3fn:
+        raise TypeError("c must be fp32; a, b must be float8_e4m3fn")
+    for t, name in ((c, "c"), (a, "a"), (b, "b")):
+        if not t.is_cuda or not t.is_contiguous() or t.dim() != 2:
+            raise TypeError(f"{name} must be a contiguous 2-D CUDA tensor")
+    m, k = a.shape
+    n, kb = b.shape
+    if kb != k or c.shape != (m, n):
+        raise ValueError(f"shape mismatch: A{tuple(a.shape)} B{tuple(b.shape)}"
+                         f" C{tuple(c.shape)}")
+    native().gemm_fp8_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                         m, n, k, _stream_handle(stream), int(xcd_swizzle))

@@ -387,3 +387,32 @@ def test_gemm_bf16_db_variant_exact():
         assert torch.equal(c, ref)
     finally:
         del os.environ["HPK_GEMM_VARIANT"]
+
+
+def test_gemm_fp8_exact_integers():
+    """fp8 e4m3 twin: {-4..4} are exactly representable in e4m3, so the
+    result must EQUAL the torch fp32 reference bitwise."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(17)
+    m, n, k = 256, 384, 512
+    a = torch.randint(-4, 5, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_fp8(c, a, b)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_fp8_identity_asymmetric():
+    from hpc_patterns_amd import ops
+
+    torch.manual_seed(5)
+    m = n = k = 128
+    a = torch.eye(m, k, device="cuda").to(torch.float8_e4m3fn)
+    b = (torch.randint(-4, 5, (n, k)).float()).to(torch.float8_e4m3fn).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_fp8(c, a, b)
+    torch.cuda.synchronize()
+    assert torch.equal(c, b.float().t())
