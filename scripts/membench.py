@@ -106,13 +106,14 @@ def main():
         a = (torch.rand(sz, sz, device=dev) * 2 - 1).to(torch.bfloat16)
         b = (torch.rand(sz, sz, device=dev) * 2 - 1).to(torch.bfloat16)
         c = torch.empty(sz, sz, dtype=torch.float32, device=dev)
-        for sw in (True, False):
-            t = time_gpu(lambda: ops.gemm_bf16(c, a, b, xcd_swizzle=sw))
-            fl = 2.0 * sz * sz * sz
-            tag = "swz" if sw else "lin"
-            print(f"gemm {sz:5d} {tag} {t*1e3:9.3f} ms  "
-                  f"{fl/t/1e12:9.1f} TFLOP/s bf16")
-        del a, b, c
+        fl = 2.0 * sz * sz * sz
+        t = time_gpu(lambda: ops.gemm_bf16(c, a, b))
+        print(f"gemm {sz:5d} bf16 {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
+        a8 = a.to(torch.float8_e4m3fn)
+        b8 = b.to(torch.float8_e4m3fn)
+        t = time_gpu(lambda: ops.gemm_fp8(c, a8, b8))
+        print(f"gemm {sz:5d} fp8  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
+        del a, b, a8, b8, c
 
 
 if __name__ == "__main__":
