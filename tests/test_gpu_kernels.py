@@ -416,3 +416,26 @@ def test_gemm_fp8_identity_asymmetric():
     ops.gemm_fp8(c, a, b)
     torch.cuda.synchronize()
     assert torch.equal(c, b.float().t())
+
+
+def test_gemm_bf16_8phase_exact_integers():
+    """The deep-pipelined 256^2 8-phase kernel (default for shapes with
+    M,N % 256 == 0 and K % 128 == 0): bitwise equality on integer payloads
+    at an eligible shape — plus the identity check at 256^2."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(29)
+    m, n, k = 512, 256, 640
+    a = torch.randint(-3, 4, (m, k), generator=g).to(torch.bfloat16).cuda()
+    b = torch.randint(-3, 4, (n, k), generator=g).to(torch.bfloat16).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(c, a, b)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+    ai = torch.eye(256, 256, device="cuda").to(torch.bfloat16)
+    bi = torch.randn(256, 256, device="cuda").to(torch.bfloat16)
+    ci = torch.empty(256, 256, dtype=torch.float32, device="cuda")
+    ops.gemm_bf16(ci, ai, bi)
+    torch.cuda.synchronize()
+    assert torch.equal(ci, bi.t().float())
