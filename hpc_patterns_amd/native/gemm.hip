@@ -570,6 +570,188 @@ __global__ __launch_bounds__(512) void k_gemm_bf16_8ph(
     }
 }
 
+// ---------------------------------------------------------------------------
+// K7-8ph-fp8: the deep-pipelined 256^2 schedule at OCP fp8 e4m3 —
+// half-tile images are 8 KiB (64 KiB LDS total -> 2 blocks/CU), one glds
+// per half per thread (tile-switch waits become vmcnt(2)), fragments are
+// 8 packed fp8 in an i64 read by ds_read_b64 through the fp8 skew.
+// ---------------------------------------------------------------------------
+constexpr int PHALF8 = 128 * 64; // fp8 half-tile image (8 KiB)
+
+__global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, int M, int N, int K,
+    int tiles_n, int nwg, int xcd_swizzle) {
+  __shared__ unsigned char lds[8 * PHALF8]; // 64 KiB -> 2 blocks/CU
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 2;       // 2 wave-rows of 128
+  const int wc = wid & 3;        // 4 wave-cols of 64
+
+  const int ntiles = K / 64;
+
+  // half ids within a tile: 0 = B cols 0-127, 1 = B cols 128-255,
+  // 2 = A rows 0-127, 3 = A rows 128-255 (also the per-tile issue order)
+  auto stage_half = [&](int tile, int half) {
+    int k0 = (tile < ntiles ? tile : ntiles - 1) * 64; // clamp = pad refetch
+    int slot = (tile & 1) * 4 + half;
+    const unsigned char* G;
+    long rbase;
+    if (half < 2) {
+      G = B;
+      rbase = bcol + half * 128;
+    } else {
+      G = A;
+      rbase = brow + (half - 2) * 128;
+    }
+    {
+      // one issue: 512 threads x 16 B = the whole 8-KiB half image
+      long o_base = (long)wid * 1024;
+      long o = lds_unskew8(o_base + (long)lane * 16);
+      int row = (int)(o >> 6);
+      int kk = (int)(o & 63);
+      const unsigned char* g = G + (rbase + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)g,
+          (__attribute__((address_space(3))) void*)(lds + (long)slot * PHALF8 +
+                                                    o_base),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[8][4] = {};
+  long af[4][2];  // current m-group (4 m) x kk (8 packed fp8)
+  long bl[2][2];  // n0-1 x kk
+  long bh[2][2];  // n2-3 x kk
+
+  const int a_half = 2 + wr;       // this wave's A half id
+  const int b_half = wc >> 1;      // this wave's B half id
+  const int bcol_in_half = (wc & 1) * 64;
+
+  auto read_a = [&](int parity, int mg) {
+    const unsigned char* sa = lds + (long)(parity * 4 + a_half) * PHALF8;
+    for (int m = 0; m < 4; ++m)
+      for (int k2 = 0; k2 < 2; ++k2) {
+        int row = mg * 64 + m * 16 + (lane & 15);
+        int kf = k2 * 32 + 8 * (lane >> 4);
+        af[m][k2] = *(const long*)(sa + lds_skew8(row * 64 + kf));
+      }
+  };
+  auto read_b = [&](int parity, int ng, long (*dst)[2]) {
+    const unsigned char* sb = lds + (long)(parity * 4 + b_half) * PHALF8;
+    for (int n = 0; n < 2; ++n)
+      for (int k2 = 0; k2 < 2; ++k2) {
+        int col = bcol_in_half + (ng * 2 + n) * 16 + (lane & 15);
+        int kf = k2 * 32 + 8 * (lane >> 4);
+        dst[n][k2] = *(const long*)(sb + lds_skew8(col * 64 + kf));
+      }
+  };
+  auto mfma16 = [&](int mg, int ng, long (*bfr)[2]) {
+    for (int m = 0; m < 4; ++m)
+      for (int n = 0; n < 2; ++n)
+        for (int k2 = 0; k2 < 2; ++k2)
+          acc[mg * 4 + m][ng * 2 + n] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              af[m][k2], bfr[n][k2], acc[mg * 4 + m][ng * 2 + n], 0, 0, 0);
+  };
+  auto phase_sync_pre = [&] {
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+  };
+  auto phase_sync_post = [&] {
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  };
+
+  // prologue: B(0)h0 h1, A(0)h0 h1, B(1)h0 h1 — the steady-state tail.
+  // The vmcnt+barrier pair publishes tile 0: every wave retires its own
+  // share of the tile's DMAs BEFORE the barrier, so after it the whole
+  // tile is LDS-visible to every reader (vmcnt is per-wave; the barrier
+  // is what turns "my DMAs done" into "all DMAs done").
+  stage_half(0, 0);
+  stage_half(0, 1);
+  stage_half(0, 2);
+  stage_half(0, 3);
+  stage_half(1, 0);
+  stage_half(1, 1);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // my tile-0 DMAs done
+  __builtin_amdgcn_s_barrier();                    // ...everyone's done
+
+  for (int t = 0; t < ntiles; t += 2) {
+    const int p0 = 0, p1 = 1; // even tile -> parity 0 slots
+    // ph0
+    stage_half(t + 1, 2);
+    read_a(p0, 0);
+    read_b(p0, 0, bl);
+    phase_sync_pre();
+    mfma16(0, 0, bl);
+    phase_sync_post();
+    // ph1
+    stage_half(t + 1, 3);
+    read_b(p0, 1, bh);
+    phase_sync_pre();
+    mfma16(0, 1, bh);
+    phase_sync_post();
+    // ph2
+    stage_half(t + 2, 0);
+    read_a(p0, 1);
+    phase_sync_pre();
+    mfma16(1, 0, bl);
+    phase_sync_post();
+    // ph3 — closing barrier also publishes tile t+1 (vmcnt before it)
+    stage_half(t + 2, 1);
+    phase_sync_pre();
+    mfma16(1, 1, bh);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // my t+1 DMAs done
+    __builtin_amdgcn_s_barrier();                    // all t+1 DMAs done
+    // ph4 — tile t+1 (parity 1)
+    stage_half(t + 2, 2);
+    read_a(p1, 0);
+    read_b(p1, 0, bl);
+    phase_sync_pre();
+    mfma16(0, 0, bl);
+    phase_sync_post();
+    // ph5
+    stage_half(t + 2, 3);
+    read_b(p1, 1, bh);
+    phase_sync_pre();
+    mfma16(0, 1, bh);
+    phase_sync_post();
+    // ph6
+    stage_half(t + 3, 0);
+    read_a(p1, 1);
+    phase_sync_pre();
+    mfma16(1, 0, bl);
+    phase_sync_post();
+    // ph7 — closing barrier also publishes tile t+2 for the next ph0
+    stage_half(t + 3, 1);
+    phase_sync_pre();
+    mfma16(1, 1, bh);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); // my t+2 DMAs done
+    __builtin_amdgcn_s_barrier();                    // all t+2 DMAs done
+  }
+
+  for (int m = 0; m < 8; ++m)
+    for (int n = 0; n < 4; ++n) {
+      long row0 = brow + wr * 128 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 64 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 } // namespace
 
 void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
@@ -578,6 +760,17 @@ void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
   if (M % BM != 0 || N % BN != 0 || K % BK != 0)
     throw std::runtime_error(
         "gemm_fp8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  const char* var = std::getenv("HPK_GEMM_VARIANT");
+  const bool ph8 = !var || std::string(var) == "8ph";
+  if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0) {
+    int tn = (int)(N / 256);
+    int n8 = (int)(M / 256) * tn;
+    hipLaunchKernelGGL(k_gemm_fp8_8ph, dim3(n8), dim3(512), 0, stream, C,
+                       (const unsigned char*)A, (const unsigned char*)B,
+                       (int)M, (int)N, (int)K, tn, n8, xcd_swizzle);
+    check_hip(hipGetLastError(), "launch_gemm_fp8_nt(8ph)");
+    return;
+  }
   int tiles_m = (int)(M / BM), tiles_n = (int)(N / BN);
   int nwg = tiles_m * tiles_n;
   hipLaunchKernelGGL((k_gemm_fp8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,

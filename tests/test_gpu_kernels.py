@@ -439,3 +439,18 @@ def test_gemm_bf16_8phase_exact_integers():
     ops.gemm_bf16(ci, ai, bi)
     torch.cuda.synchronize()
     assert torch.equal(ci, bi.t().float())
+
+
+def test_gemm_fp8_8phase_exact_integers():
+    """fp8 8-phase default path (M,N%256, K%128): bitwise equality."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(37)
+    m, n, k = 256, 512, 640
+    a = torch.randint(-3, 4, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-3, 4, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_fp8(c, a, b)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
