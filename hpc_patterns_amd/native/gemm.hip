@@ -44,25 +44,30 @@ constexpr int BM = 128, BN = 128, BK = 64;
 // one K-step's staging: A[128][64] + B[128][64] bf16 = 32 KiB
 constexpr int TILE_HALF = BM * BK;            // elements per operand tile
 
-// Cyclic-skew LDS layout: linear [128][64]-bf16 rows are 128 B, so the
-// 16 rows a ds_read_b128 lane group touches land on only two 4-bank
-// windows (8-way conflict; st_16x32 XOR halves it to 2-way — measured
-// SQ_LDS_BANK_CONFLICT still 0.5x of IDX_ACTIVE). Rotating each row's
-// 64-element K range by 8 elements per row PAIR puts all 16 rows of a
-// lane group on 16 DISTINCT 4-dword bank windows:
-//   LDS(row, k) = (row, (k + 8*((row>>1)&7)) & 63)
-// The 8-element granule keeps every 16-B glds chunk and every 8-element
+// Cyclic-skew LDS layout: linear [128][64]-bf16 rows are 128 B, so
+// ds_read_b128 fragment reads pile multiple lanes onto the same 4-dword
+// bank window. gfx950's b128 lane groups are NOT contiguous 16-lane
+// blocks — they mix rows AND kfrag halves (e.g. {0-3,12-15,20-27}:
+// rows {0-3,12-15} at kfrag q and rows {4-11} at kfrag q+1,
+// MI355X_MICROARCH.md LDS table) — so the shift must solve the mixed
+// sets: rotating each row's K range by 16 elements per (row>>1)&3 class,
+//   LDS(row, k) = (row, (k + 16*((row>>1)&3)) & 63)
+// gives every row class an EVEN 4-dword window slot and the interleaved
+// kfrag half the odd slots: all 16 lanes of each true lane group land on
+// 16 distinct windows (verified by enumeration; a naive per-row-pair
+// 8-element shift measured SQ_LDS_BANK_CONFLICT = 0.5x IDX_ACTIVE —
+// 2-way residual — precisely because of the group mixing).
+// The 16-element granule keeps every 16-B glds chunk and every 8-element
 // fragment read contiguous, so the staging pre-applies the inverse on the
 // GLOBAL source chunk address while the LDS write stays lane-linear (the
-// global_load_lds requirement); the skew costs only slightly shuffled
-// global fetch order within each 128-B row.
+// global_load_lds requirement).
 __device__ __forceinline__ long lds_skew(long e) { // tile elem -> LDS slot
   long row = e >> 6, k = e & 63;
-  return (row << 6) | ((k + 8 * ((row >> 1) & 7)) & 63);
+  return (row << 6) | ((k + 16 * ((row >> 1) & 3)) & 63);
 }
 __device__ __forceinline__ long lds_unskew(long y) { // LDS slot -> tile elem
   long row = y >> 6, k = y & 63;
-  return (row << 6) | ((k - 8 * ((row >> 1) & 7)) & 63);
+  return (row << 6) | ((k - 16 * ((row >> 1) & 3)) & 63);
 }
 
 // Wave-grid decomposition is a template knob: <2,2> = 4 waves of 64x64
