@@ -274,12 +274,19 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt_db(
 // ---------------------------------------------------------------------------
 // K7-fp8: same GEMM at OCP fp8 e4m3 (__builtin_amdgcn_mfma_f32_16x16x32_
 // fp8_fp8 runs at the bf16 MFMA rate but the staging traffic halves).
-// A/B fragments are 8 packed fp8 in one i64 (2 VGPRs), read from LDS by
-// ds_read_b64. The [128][64]-byte rows give ds_read_b64 lane groups of 32
-// a 4-way bank conflict in linear layout; shifting each row's K range by
-// 16 elements per 4-row class — LDS(row,k) = (row, (k + 16*((row>>2)&3))
-// & 63) — plus the kfrag offset spreads all 32 lanes of a group over 32
-// distinct 2-dword windows (conflict-free).
+// A/B fragments are 8 packed fp8 in one i64 (2 VGPRs); LDS(row,k) =
+// (row, (k + 16*((row>>2)&3)) & 63) spreads the fragment reads.
+//
+// Known 2-way LDS-conflict residual (measured SQ_LDS_BANK_CONFLICT =
+// 0.5x IDX_ACTIVE, ~6% of wall): the compiler emits the 8-byte fragment
+// reads as paired ds_read2st64_b64 whose banking is (a/4) mod 32 with
+// 16-contiguous-lane groups; a conflict-FREE assignment there needs
+// 8-byte K-rotations, but an 8-byte shift makes some 16-byte
+// global_load_lds chunks wrap the 64-byte K range (source becomes
+// non-contiguous), and row padding breaks glds lane-linearity — so
+// 16-byte granularity (2-way) is the floor for this staging scheme.
+// bf16's 16-bit elements dodge this: its 16-element rotation is one
+// b128 window and reaches zero conflicts (see lds_skew above).
 // ---------------------------------------------------------------------------
 __device__ __forceinline__ long lds_skew8(long e) { // tile elem -> LDS slot
   long row = e >> 6, k = e & 63;
@@ -349,11 +356,13 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_fp8_nt(
       long afrag[MREP], bfrag[NREP];
       for (int m = 0; m < MREP; ++m) {
         int row = wr * WTM + m * 16 + (lane & 15);
-        afrag[m] = *(const long*)(la + lds_skew8(row * BK + kfrag));
+        afrag[m] = *(const long*)__builtin_assume_aligned(
+            la + lds_skew8(row * BK + kfrag), 8);
       }
       for (int n = 0; n < NREP; ++n) {
         int col = wc * WTN + n * 16 + (lane & 15);
-        bfrag[n] = *(const long*)(lb + lds_skew8(col * BK + kfrag));
+        bfrag[n] = *(const long*)__builtin_assume_aligned(
+            lb + lds_skew8(col * BK + kfrag), 8);
       }
       for (int m = 0; m < MREP; ++m)
         for (int n = 0; n < NREP; ++n)
@@ -649,7 +658,8 @@ __global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
       for (int k2 = 0; k2 < 2; ++k2) {
         int row = mg * 64 + m * 16 + (lane & 15);
         int kf = k2 * 32 + 8 * (lane >> 4);
-        af[m][k2] = *(const long*)(sa + lds_skew8(row * 64 + kf));
+        af[m][k2] = *(const long*)__builtin_assume_aligned(
+            sa + lds_skew8(row * 64 + kf), 8);
       }
   };
   auto read_b = [&](int parity, int ng, long (*dst)[2]) {
@@ -658,7 +668,8 @@ __global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
       for (int k2 = 0; k2 < 2; ++k2) {
         int col = bcol_in_half + (ng * 2 + n) * 16 + (lane & 15);
         int kf = k2 * 32 + 8 * (lane >> 4);
-        dst[n][k2] = *(const long*)(sb + lds_skew8(col * 64 + kf));
+        dst[n][k2] = *(const long*)__builtin_assume_aligned(
+            sb + lds_skew8(col * 64 + kf), 8);
       }
   };
   auto mfma16 = [&](int mg, int ng, long (*bfr)[2]) {
