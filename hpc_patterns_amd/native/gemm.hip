@@ -768,7 +768,138 @@ __global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
     }
 }
 
+// ---------------------------------------------------------------------------
+// K7-mxfp8: block-scaled MX-fp8 GEMM — __builtin_amdgcn_mfma_scale_f32_
+// 16x16x128_f8f6f4 runs at TWICE the bf16/fp8 MFMA rate (the only path to
+// the low-precision peaks on gfx950) with HW-fused dequant: every
+// 32-element K-block of A and B carries one e8m0 scale (byte; 127 = 1.0),
+// exactly the OCP MX-FP8 format.
+//   C[M,N] fp32 = (A .* 2^(As-127)) x (B .* 2^(Bs-127))^T
+// A/B fp8 e4m3 [M][K]/[N][K]; As/Bs uint8 [M][K/32]/[N][K/32].
+// Fragment mapping verified on hardware by bin/mx_probe: lane L holds 32
+// contiguous k at 32*(L>>4), row/col L&15; C/D standard; the scale VGPR's
+// byte 0 (opsel 0) scales the lane's block.
+// Structure: plain two-barrier K-loop, BK = 128 (one MFMA per fragment
+// pair), 8 waves of 64x32; scales ride ordinary loads + ds_write into the
+// same LDS block (the __syncthreads drain makes them free here — in the
+// deep-pipelined schedule they would puncture the counted-vmcnt pipeline,
+// the documented trap 4(b)).
+// LDS skew (b128 reads at 16-B granule in a [128][128]-byte image): the
+// +4-invariant window family w = {0,1,4,5,8,9,12,13}: rotation
+// k' = (k + 16*((row>>1)&5)) & 127 makes ALL FOUR true ds_read_b128 lane
+// groups conflict-free (enumerated like lds_skew above).
+// ---------------------------------------------------------------------------
+constexpr int MXK = 128; // K-step
+__device__ __forceinline__ long mx_skew(long e) { // byte index in [128][128]
+  long row = e >> 7, k = e & 127;
+  return (row << 7) | ((k + 16 * ((row >> 1) & 5)) & 127);
+}
+__device__ __forceinline__ long mx_unskew(long y) {
+  long row = y >> 7, k = y & 127;
+  return (row << 7) | ((k - 16 * ((row >> 1) & 5)) & 127);
+}
+
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+
+__global__ __launch_bounds__(512) void k_gemm_mxfp8_nt(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle) {
+  constexpr int MREP = 4, NREP = 2; // 8 waves as 2x4, 64x32 per wave
+  // [A data 16K][B data 16K][A scales 512][B scales 512]
+  __shared__ unsigned char lds[2 * 128 * MXK + 2 * 512];
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * 128;
+  const long bcol = (long)(wg % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+  const int ks = K / 32; // scale row stride
+
+  unsigned char* sA = lds + 2 * 128 * MXK;
+  unsigned char* sB = sA + 512;
+  f32x4 acc[MREP][NREP] = {};
+
+  for (int k0 = 0; k0 < K; k0 += MXK) {
+    __syncthreads();
+    // data: [128][128] bytes per operand = 512 threads x 16 B x 2 issues
+    for (int issue = 0; issue < 2; ++issue) {
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = mx_unskew(o_base + (long)lane * 16);
+      int row = (int)(o >> 7);
+      int kk = (int)(o & 127);
+      const unsigned char* ga = A + (brow + row) * (long)K + k0 + kk;
+      const unsigned char* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(lds + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(lds + 128 * MXK + o_base),
+          16, 0, 0);
+    }
+    // scales: 128 rows x 4 k-blocks per operand; thread t stages one byte
+    // of each (ordinary load + ds_write — the barrier drains them anyway)
+    {
+      int row = tid >> 2, kb = tid & 3;
+      sA[tid] = As[(brow + row) * (long)ks + k0 / 32 + kb];
+      sB[tid] = Bs[(bcol + row) * (long)ks + k0 / 32 + kb];
+    }
+    __syncthreads();
+
+    const int kb = lane >> 4; // this lane's 32-elem block within the step
+    i32x8 afrag[MREP];
+    int asc[MREP];
+    for (int m = 0; m < MREP; ++m) {
+      int row = wr * 64 + m * 16 + (lane & 15);
+      afrag[m] = *(const i32x8*)__builtin_assume_aligned(
+          lds + mx_skew((long)row * MXK + 32 * kb), 16);
+      asc[m] = sA[row * 4 + kb];
+    }
+    for (int n = 0; n < NREP; ++n) {
+      int col = wc * 32 + n * 16 + (lane & 15);
+      i32x8 bfrag = *(const i32x8*)__builtin_assume_aligned(
+          lds + 128 * MXK + mx_skew((long)col * MXK + 32 * kb), 16);
+      int bsc = sB[col * 4 + kb];
+      for (int m = 0; m < MREP; ++m)
+        acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            afrag[m], bfrag, acc[m][n], 0, 0, 0, asc[m], 0, bsc);
+    }
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * 64 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 32 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 } // namespace
+
+void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
+                          const void* As, const void* Bs, long M, long N,
+                          long K, hipStream_t stream, int xcd_swizzle) {
+  if (M % 128 != 0 || N % 128 != 0 || K % 128 != 0)
+    throw std::runtime_error("gemm_mxfp8_nt requires M,N,K % 128 == 0");
+  int tiles_n = (int)(N / 128);
+  int nwg = (int)(M / 128) * tiles_n;
+  hipLaunchKernelGGL(k_gemm_mxfp8_nt, dim3(nwg), dim3(512), 0, stream, C,
+                     (const unsigned char*)A, (const unsigned char*)B,
+                     (const unsigned char*)As, (const unsigned char*)Bs,
+                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+  check_hip(hipGetLastError(), "launch_gemm_mxfp8_nt");
+}
 
 void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
                         long N, long K, hipStream_t stream,

@@ -155,3 +155,35 @@ def gemm_fp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
                          f" C{tuple(c.shape)}")
     native().gemm_fp8_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
                          m, n, k, _stream_handle(stream), int(xcd_swizzle))
+
+
+def gemm_mxfp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
+               a_scale: torch.Tensor, b_scale: torch.Tensor,
+               stream=None, xcd_swizzle: bool = False) -> None:
+    """K7-mx: block-scaled MX-fp8 GEMM at 2x the bf16 MFMA rate.
+
+    C[M,N] fp32 = (A * 2^(As-127)) @ (B * 2^(Bs-127))^T — A/B are
+    float8_e4m3fn [M,K]/[N,K]; a_scale/b_scale are uint8 e8m0 exponents
+    with ONE scale per 32-element K-block ([M,K//32] / [N,K//32]; 127 =
+    scale 1.0) — the OCP MX-FP8 format, dequantized in hardware by
+    mfma_scale_f32_16x16x128_f8f6f4. Requires M,N,K multiples of 128.
+    """
+    if c.dtype != torch.float32 or a.dtype != torch.float8_e4m3fn \
+            or b.dtype != torch.float8_e4m3fn:
+        raise TypeError("c must be fp32; a, b must be float8_e4m3fn")
+    if a_scale.dtype != torch.uint8 or b_scale.dtype != torch.uint8:
+        raise TypeError("scales must be uint8 (e8m0 exponents)")
+    for t, name in ((c, "c"), (a, "a"), (b, "b"),
+                    (a_scale, "a_scale"), (b_scale, "b_scale")):
+        if not t.is_cuda or not t.is_contiguous() or t.dim() != 2:
+            raise TypeError(f"{name} must be a contiguous 2-D CUDA tensor")
+    m, k = a.shape
+    n, kb = b.shape
+    if kb != k or c.shape != (m, n):
+        raise ValueError(f"shape mismatch: A{tuple(a.shape)} B{tuple(b.shape)}"
+                         f" C{tuple(c.shape)}")
+    if a_scale.shape != (m, k // 32) or b_scale.shape != (n, k // 32):
+        raise ValueError("scales must be [rows, K//32]")
+    native().gemm_mxfp8_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                           a_scale.data_ptr(), b_scale.data_ptr(),
+                           m, n, k, _stream_handle(stream), int(xcd_swizzle))

@@ -454,3 +454,61 @@ def test_gemm_fp8_8phase_exact_integers():
     ref = torch.matmul(a.float(), b.float().t())
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def _mx_ref(a, b, a_scale, b_scale):
+    """fp32 reference of the MX semantics: dequant per 32-block, matmul."""
+    af = a.float() * (2.0 ** (a_scale.float() - 127)).repeat_interleave(32, 1)
+    bf = b.float() * (2.0 ** (b_scale.float() - 127)).repeat_interleave(32, 1)
+    return torch.matmul(af, bf.t())
+
+
+def test_gemm_mxfp8_unit_scales_exact():
+    """All scales 127 (=1.0): must equal the plain-fp8 exact result."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(41)
+    m, n, k = 128, 256, 384
+    a = torch.randint(-4, 5, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp8(c, a, b, sa, sb)
+    ref = torch.matmul(a.float(), b.float().t())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_mxfp8_power_of_two_scales_exact():
+    """Random per-block power-of-2 scales: products stay exact in fp32, so
+    the HW-dequant result must EQUAL the fp32 reference bitwise."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(43)
+    m, n, k = 256, 128, 512
+    a = torch.randint(-3, 4, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-3, 4, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.uint8).cuda()
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.uint8).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp8(c, a, b, sa, sb)
+    ref = _mx_ref(a, b, sa, sb)
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_mxfp8_identity():
+    from hpc_patterns_amd import ops
+
+    m = n = k = 128
+    a = torch.eye(m, k, device="cuda").to(torch.float8_e4m3fn)
+    b = torch.randint(-4, 5, (n, k)).float().to(torch.float8_e4m3fn).cuda()
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp8(c, a, b, sa, sb)
+    torch.cuda.synchronize()
+    assert torch.equal(c, b.float().t())
