@@ -512,3 +512,23 @@ def test_gemm_mxfp8_identity():
     ops.gemm_mxfp8(c, a, b, sa, sb)
     torch.cuda.synchronize()
     assert torch.equal(c, b.float().t())
+
+
+def test_gemm_mxfp8_large_square_exact():
+    """256-divisible shape with random power-of-2 scales (covers the tile
+    shapes the removed deep-pipelined variant used to take)."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(47)
+    m, n, k = 256, 256, 512
+    a = torch.randint(-3, 4, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-3, 4, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.uint8).cuda()
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.uint8).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp8(c, a, b, sa, sb)
+    ref = _mx_ref(a, b, sa, sb)
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
