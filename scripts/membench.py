@@ -113,7 +113,10 @@ def main():
         b8 = b.to(torch.float8_e4m3fn)
         t = time_gpu(lambda: ops.gemm_fp8(c, a8, b8))
         print(f"gemm {sz:5d} fp8  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
-        del a, b, a8, b8, c
+        s1 = torch.full((sz, sz // 32), 127, dtype=torch.uint8, device=dev)
+        t = time_gpu(lambda: ops.gemm_mxfp8(c, a8, b8, s1, s1))
+        print(f"gemm {sz:5d} mx8  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
+        del a, b, a8, b8, s1, c
 
 
 if __name__ == "__main__":
