@@ -1,29 +1,36 @@
-// gemm.hip — K7: LDS-tiled bf16 MFMA GEMM (beyond-parity showcase).
+// gemm.hip — K7: the LDS-tiled MFMA GEMM family (beyond-parity showcase).
 //
 // The reference suite's device kernels (K1-K6) are streaming/elementwise;
-// this kernel demonstrates the OTHER half of the CDNA4 execution model the
-// suite is designed around: matrix cores fed through LDS.
-//   C[M,N] (fp32) = A[M,K] (bf16, row-major) x B[N,K]^T (bf16, row-major)
-// i.e. an "NT" GEMM — both operands K-contiguous, the natural
-// v_mfma_f32_16x16x32_bf16 feeding order.
+// these kernels demonstrate the OTHER half of the CDNA4 execution model
+// the suite is designed around: matrix cores fed through LDS. All compute
+//   C[M,N] (fp32) = A[M,K] x B[N,K]^T  ("NT": both operands K-contiguous,
+// the natural MFMA feeding order), fp32 accumulate, verified BITWISE
+// against torch fp32 matmul on exactly-representable payloads
+// (tests/test_gpu_kernels.py) and race-screened.
 //
-// Structure (the plain-HIP two-barrier K-loop):
-//   - 128x128 output tile per 256-thread workgroup (4 waves, 2x2 wave
-//     grid, each wave owns a 64x64 sub-tile as 4x4 MFMA fragments).
-//   - K-step 64: A-tile [128][64] and B-tile [128][64] staged into ONE
-//     32 KiB LDS buffer per K-step via __builtin_amdgcn_global_load_lds
-//     (16-byte direct-to-LDS DMA; the LDS image is lane-linear by
-//     construction, which that instruction requires).
-//   - two barriers per K-step: [sync] stage [sync] 16 ds_read_b128 + 32
-//     chained MFMAs per wave.
-//   - bijective XCD-aware workgroup swizzle so consecutive XCDs see
-//     neighbouring C tiles (L2 locality when HBM-bound).
-// Fragment mappings (16x16x32 bf16): A/B lane L holds 8 contiguous K
+// Members (measured on MI355X, random operands, profiles/gemm_showcase_r2):
+//   k_gemm_bf16_nt<2,2|2,4>   plain two-barrier 128^2 tile; the 8-wave
+//                             64x32 decomposition wins (76 VGPR -> 6
+//                             waves/SIMD): 840/935 TF at 4096^3/8192^3
+//   k_gemm_bf16_nt_db         + double-buffered LDS, raw barriers,
+//                             counted vmcnt (917/879 TF)
+//   k_gemm_bf16_8ph           256^2 deep pipeline, 8 phases, two K-tiles
+//                             in 8 rotating LDS half-slots, vmcnt(4) only
+//                             at tile switches — DEFAULT for eligible
+//                             shapes: 968/1085 TF, zero LDS bank
+//                             conflicts (true-lane-group cyclic skew)
+//   k_gemm_fp8_nt / _8ph      OCP fp8 e4m3 twins (same MFMA rate, half
+//                             the staging bytes): up to 1264 TF
+//   k_gemm_mxfp8_nt           block-scaled OCP MX-fp8 via
+//                             mfma_scale_f32_16x16x128_f8f6f4 (HW-fused
+//                             e8m0 dequant; scale lane layout
+//                             reverse-engineered on hardware —
+//                             scripts/probes/): 1216 TF
+//
+// Fragment mappings (16x16x32 bf16/fp8): A/B lane L holds 8 contiguous K
 // elements at k = 8*(L>>4), row/col = L&15; C/D lane L reg r holds
-// row = 4*(L>>4)+r, col = L&15.
-//
-// Numerics: fp32 accumulate; verified against torch fp32 matmul with
-// exactly-representable integer payloads (tests/test_gpu_kernels.py).
+// row = 4*(L>>4)+r, col = L&15 (identical across shapes/dtypes). The MX
+// instruction's operand/scale layout is documented at its kernel.
 
 #include "include/hpk.h"
 
