@@ -76,6 +76,16 @@ PYBIND11_MODULE(_hpk, m) {
         },
         py::arg("c"), py::arg("a"), py::arg("b"), py::arg("m"), py::arg("n"),
         py::arg("k"), py::arg("stream") = 0, py::arg("xcd_swizzle") = 0);
+  m.def("gemm_i8_nt",
+        [](uintptr_t c, uintptr_t a, uintptr_t b, long m, long n, long k,
+           uintptr_t stream, int xcd_swizzle) {
+          hpk::launch_gemm_i8_nt(reinterpret_cast<int*>(c),
+                                 reinterpret_cast<const void*>(a),
+                                 reinterpret_cast<const void*>(b), m, n, k,
+                                 as_stream(stream), xcd_swizzle);
+        },
+        py::arg("c"), py::arg("a"), py::arg("b"), py::arg("m"), py::arg("n"),
+        py::arg("k"), py::arg("stream") = 0, py::arg("xcd_swizzle") = 0);
   m.def("gemm_mxfp8_nt",
         [](uintptr_t c, uintptr_t a, uintptr_t b, uintptr_t as, uintptr_t bs,
            long m, long n, long k, uintptr_t stream, int xcd_swizzle) {

@@ -916,6 +916,105 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt(
     }
 }
 
+// ---------------------------------------------------------------------------
+// K7-i8: integer GEMM — mfma_i32_16x16x64_i8 runs at ~2x the bf16 rate
+// with exact int32 accumulation (the quantized-inference dtype).
+//   C[M,N] (int32) = A[M,K] (int8) x B[N,K]^T (int8)
+// K=64 per MFMA: each lane holds 16 contiguous int8 = one ds_read_b128.
+// No scales, so the lane->k assignment only needs A/B consistency
+// (matmul is K-permutation-invariant) — the natural k = 16*(lane>>4)
+// order is used. LDS images are [128][64] bytes; the b128 lane-group skew
+// for this geometry is a 32-byte rotation per (row>>3)&1 class (same
+// derivation style as lds_skew: row classes on even 4-dword windows, the
+// interleaved kfrag half on odd ones).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ long i8_skew(long e) {
+  long row = e >> 6, k = e & 63;
+  return (row << 6) | ((k + 32 * ((row >> 3) & 1)) & 63);
+}
+__device__ __forceinline__ long i8_unskew(long y) {
+  long row = y >> 6, k = y & 63;
+  return (row << 6) | ((k - 32 * ((row >> 3) & 1)) & 63);
+}
+
+typedef __attribute__((ext_vector_type(4))) int i32x4v;
+
+template <int WAVES_M, int WAVES_N>
+__global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_i8_nt(
+    int* __restrict__ C, const signed char* __restrict__ A,
+    const signed char* __restrict__ B, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle) {
+  constexpr int THREADS = WAVES_M * WAVES_N * 64;
+  constexpr int MREP = BM / (WAVES_M * 16);
+  constexpr int NREP = BN / (WAVES_N * 16);
+  constexpr int WTM = BM / WAVES_M;
+  constexpr int WTN = BN / WAVES_N;
+  __shared__ signed char lds[2 * TILE_HALF]; // bytes: [A tile][B tile]
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * BM;
+  const long bcol = (long)(wg % tiles_n) * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WAVES_N;
+  const int wc = wid % WAVES_N;
+
+  constexpr long bytes_per_issue = (long)THREADS * 16;
+  constexpr int ISSUES = TILE_HALF / (THREADS * 16);
+  static_assert(ISSUES >= 1, "tile too small");
+  i32x4v acc[MREP][NREP] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    for (int issue = 0; issue < ISSUES; ++issue) {
+      long o_base = (long)issue * bytes_per_issue + (long)wid * (64 * 16);
+      long o = i8_unskew(o_base + (long)lane * 16);
+      int row = (int)(o >> 6);
+      int kk = (int)(o & 63);
+      const signed char* ga = A + (brow + row) * (long)K + k0 + kk;
+      const signed char* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(lds + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(lds + TILE_HALF + o_base),
+          16, 0, 0);
+    }
+    __syncthreads();
+
+    const int kfrag = 16 * (lane >> 4);
+    i32x4v afrag[MREP];
+    for (int m = 0; m < MREP; ++m) {
+      int row = wr * WTM + m * 16 + (lane & 15);
+      afrag[m] = *(const i32x4v*)__builtin_assume_aligned(
+          lds + i8_skew(row * BK + kfrag), 16);
+    }
+    for (int n = 0; n < NREP; ++n) {
+      int col = wc * WTN + n * 16 + (lane & 15);
+      i32x4v bfrag = *(const i32x4v*)__builtin_assume_aligned(
+          lds + TILE_HALF + i8_skew(col * BK + kfrag), 16);
+      for (int m = 0; m < MREP; ++m)
+        acc[m][n] = __builtin_amdgcn_mfma_i32_16x16x64_i8(afrag[m], bfrag,
+                                                          acc[m][n], 0, 0, 0);
+    }
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * WTM + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * WTN + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 // NOTE (r2): a deep-pipelined 8-phase MX variant was built and measured,
 // then REMOVED: MX fragments are 8 VGPRs per operand (32 bytes) and the
 // 256^2 schedule's 128-VGPR accumulator plus the scale plumbing exceeded
@@ -926,6 +1025,19 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt(
 // scaled instruction at this tile geometry.
 
 } // namespace
+
+void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
+                       long N, long K, hipStream_t stream, int xcd_swizzle) {
+  if (M % BM != 0 || N % BN != 0 || K % BK != 0)
+    throw std::runtime_error(
+        "gemm_i8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  int tiles_n = (int)(N / BN);
+  int nwg = (int)(M / BM) * tiles_n;
+  hipLaunchKernelGGL((k_gemm_i8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,
+                     C, (const signed char*)A, (const signed char*)B, (int)M,
+                     (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+  check_hip(hipGetLastError(), "launch_gemm_i8_nt");
+}
 
 void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,

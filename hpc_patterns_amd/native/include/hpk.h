@@ -77,6 +77,11 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
 void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
                         long N, long K, hipStream_t stream,
                         int xcd_swizzle = 0);
+// int8 GEMM (mfma_i32_16x16x64_i8, ~2x the bf16 rate, exact int32
+// accumulation): C[M,N] int32 = A[M,K] x B[N,K]^T, int8 operands.
+void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
+                       long N, long K, hipStream_t stream,
+                       int xcd_swizzle = 0);
 // Block-scaled MX-fp8 (mfma_scale_f32_16x16x128_f8f6f4, 2x the bf16
 // rate): C = (A .* 2^(As-127)) x (B .* 2^(Bs-127))^T with one e8m0 scale
 // byte per 32-element K-block (As: [M][K/32], Bs: [N][K/32], uint8).

@@ -532,3 +532,31 @@ def test_gemm_mxfp8_large_square_exact():
     ref = _mx_ref(a, b, sa, sb)
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_i8_exact():
+    """int8 GEMM: int32 accumulate is exact by construction — full-range
+    payloads must EQUAL the int64 torch reference."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(53)
+    m, n, k = 256, 384, 512
+    a = torch.randint(-128, 128, (m, k), generator=g, dtype=torch.int8).cuda()
+    b = torch.randint(-128, 128, (n, k), generator=g, dtype=torch.int8).cuda()
+    c = torch.empty(m, n, dtype=torch.int32, device="cuda")
+    ops.gemm_i8(c, a, b)
+    ref = torch.matmul(a.long(), b.long().t()).to(torch.int32)
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_i8_identity():
+    from hpc_patterns_amd import ops
+
+    m = n = k = 128
+    a = torch.eye(m, k).to(torch.int8).cuda()
+    b = torch.randint(-5, 6, (n, k), dtype=torch.int8).cuda()
+    c = torch.empty(m, n, dtype=torch.int32, device="cuda")
+    ops.gemm_i8(c, a, b)
+    torch.cuda.synchronize()
+    assert torch.equal(c, b.int().t())
