@@ -545,7 +545,8 @@ def test_gemm_i8_exact():
     b = torch.randint(-128, 128, (n, k), generator=g, dtype=torch.int8).cuda()
     c = torch.empty(m, n, dtype=torch.int32, device="cuda")
     ops.gemm_i8(c, a, b)
-    ref = torch.matmul(a.long(), b.long().t()).to(torch.int32)
+    # torch has no int CUDA matmul — exact reference on host
+    ref = torch.matmul(a.cpu().long(), b.cpu().long().t()).to(torch.int32).cuda()
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
 
