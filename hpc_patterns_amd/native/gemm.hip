@@ -1026,11 +1026,216 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_i8_nt(
 
 } // namespace
 
+// ---------------------------------------------------------------------------
+// K7-i8-8ph: the deep-pipelined 8-phase schedule at int8 — [128][128]-byte
+// half images (16 KiB, same slot geometry and vmcnt(4) schedule as the
+// bf16 8-phase kernel), 16-byte b128 fragments (4 VGPRs each, the
+// LIGHTEST of the family), two K=64 MFMAs per fragment pair per K-tile.
+// Skew: 32-byte rotation per (row>>1)&3 class — the evens window family
+// for this geometry's +-1-unit lane-group mixing.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ long i8s_skew(long e) {
+  long row = e >> 7, k = e & 127;
+  return (row << 7) | ((k + 32 * ((row >> 1) & 3)) & 127);
+}
+__device__ __forceinline__ long i8s_unskew(long y) {
+  long row = y >> 7, k = y & 127;
+  return (row << 7) | ((k - 32 * ((row >> 1) & 3)) & 127);
+}
+
+constexpr int PH8I = 128 * 128; // i8 half-tile image bytes (16 KiB)
+
+__global__ __launch_bounds__(512) void k_gemm_i8_8ph(
+    int* __restrict__ C, const signed char* __restrict__ A,
+    const signed char* __restrict__ B, int M, int N, int K,
+    int tiles_n, int nwg, int xcd_swizzle) {
+  __shared__ signed char lds[8 * PH8I]; // 128 KiB
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 2;       // 2 wave-rows of 128
+  const int wc = wid & 3;        // 4 wave-cols of 64
+
+  const int ntiles = K / 128;
+
+  // half ids within a tile: 0 = B cols 0-127, 1 = B cols 128-255,
+  // 2 = A rows 0-127, 3 = A rows 128-255 (also the per-tile issue order)
+  auto stage_half = [&](int tile, int half) {
+    int k0 = (tile < ntiles ? tile : ntiles - 1) * 128; // clamp = pad refetch
+    int slot = (tile & 1) * 4 + half;
+    const signed char* G;
+    long rbase;
+    if (half < 2) {
+      G = B;
+      rbase = bcol + half * 128;
+    } else {
+      G = A;
+      rbase = brow + (half - 2) * 128;
+    }
+    for (int issue = 0; issue < 2; ++issue) {
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = i8s_unskew(o_base + (long)lane * 16);
+      int row = (int)(o >> 7);
+      int kk = (int)(o & 127);
+      const signed char* g = G + (rbase + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)g,
+          (__attribute__((address_space(3))) void*)(lds + (long)slot * PH8I +
+                                                    o_base),
+          16, 0, 0);
+    }
+  };
+
+  i32x4v acc[8][4] = {};
+  i32x4v af[4][2];  // current m-group (4 m) x kk
+  i32x4v bl[2][2];  // n0-1 x kk
+  i32x4v bh[2][2];  // n2-3 x kk
+
+  const int a_half = 2 + wr;       // this wave's A half id
+  const int b_half = wc >> 1;      // this wave's B half id
+  const int bcol_in_half = (wc & 1) * 64;
+
+  auto read_a = [&](int parity, int mg) {
+    const signed char* sa = lds + (long)(parity * 4 + a_half) * PH8I;
+    for (int m = 0; m < 4; ++m)
+      for (int k2 = 0; k2 < 2; ++k2) {
+        int row = mg * 64 + m * 16 + (lane & 15);
+        int kf = k2 * 64 + 16 * (lane >> 4);
+        af[m][k2] = *(const i32x4v*)__builtin_assume_aligned(
+            sa + i8s_skew(row * 128 + kf), 16);
+      }
+  };
+  auto read_b = [&](int parity, int ng, i32x4v (*dst)[2]) {
+    const signed char* sb = lds + (long)(parity * 4 + b_half) * PH8I;
+    for (int n = 0; n < 2; ++n)
+      for (int k2 = 0; k2 < 2; ++k2) {
+        int col = bcol_in_half + (ng * 2 + n) * 16 + (lane & 15);
+        int kf = k2 * 64 + 16 * (lane >> 4);
+        dst[n][k2] = *(const i32x4v*)__builtin_assume_aligned(
+            sb + i8s_skew(col * 128 + kf), 16);
+      }
+  };
+  auto mfma16 = [&](int mg, int ng, i32x4v (*bfr)[2]) {
+    for (int m = 0; m < 4; ++m)
+      for (int n = 0; n < 2; ++n)
+        for (int k2 = 0; k2 < 2; ++k2)
+          acc[mg * 4 + m][ng * 2 + n] = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+              af[m][k2], bfr[n][k2], acc[mg * 4 + m][ng * 2 + n], 0, 0, 0);
+  };
+  auto phase_sync_pre = [&] {
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+  };
+  auto phase_sync_post = [&] {
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  };
+
+  // prologue: B(0)h0 h1, A(0)h0 h1, B(1)h0 h1 — the steady-state tail.
+  // The vmcnt+barrier pair publishes tile 0: every wave retires its own
+  // share of the tile's DMAs BEFORE the barrier, so after it the whole
+  // tile is LDS-visible to every reader (vmcnt is per-wave; the barrier
+  // is what turns "my DMAs done" into "all DMAs done").
+  stage_half(0, 0);
+  stage_half(0, 1);
+  stage_half(0, 2);
+  stage_half(0, 3);
+  stage_half(1, 0);
+  stage_half(1, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); // my tile-0 DMAs done
+  __builtin_amdgcn_s_barrier();                    // ...everyone's done
+
+  for (int t = 0; t < ntiles; t += 2) {
+    const int p0 = 0, p1 = 1; // even tile -> parity 0 slots
+    // ph0
+    stage_half(t + 1, 2);
+    read_a(p0, 0);
+    read_b(p0, 0, bl);
+    phase_sync_pre();
+    mfma16(0, 0, bl);
+    phase_sync_post();
+    // ph1
+    stage_half(t + 1, 3);
+    read_b(p0, 1, bh);
+    phase_sync_pre();
+    mfma16(0, 1, bh);
+    phase_sync_post();
+    // ph2
+    stage_half(t + 2, 0);
+    read_a(p0, 1);
+    phase_sync_pre();
+    mfma16(1, 0, bl);
+    phase_sync_post();
+    // ph3 — closing barrier also publishes tile t+1 (vmcnt before it)
+    stage_half(t + 2, 1);
+    phase_sync_pre();
+    mfma16(1, 1, bh);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); // my t+1 DMAs done
+    __builtin_amdgcn_s_barrier();                    // all t+1 DMAs done
+    // ph4 — tile t+1 (parity 1)
+    stage_half(t + 2, 2);
+    read_a(p1, 0);
+    read_b(p1, 0, bl);
+    phase_sync_pre();
+    mfma16(0, 0, bl);
+    phase_sync_post();
+    // ph5
+    stage_half(t + 2, 3);
+    read_b(p1, 1, bh);
+    phase_sync_pre();
+    mfma16(0, 1, bh);
+    phase_sync_post();
+    // ph6
+    stage_half(t + 3, 0);
+    read_a(p1, 1);
+    phase_sync_pre();
+    mfma16(1, 0, bl);
+    phase_sync_post();
+    // ph7 — closing barrier also publishes tile t+2 for the next ph0
+    stage_half(t + 3, 1);
+    phase_sync_pre();
+    mfma16(1, 1, bh);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); // my t+2 DMAs done
+    __builtin_amdgcn_s_barrier();                    // all t+2 DMAs done
+  }
+
+  for (int m = 0; m < 8; ++m)
+    for (int n = 0; n < 4; ++n) {
+      long row0 = brow + wr * 128 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 64 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
                        long N, long K, hipStream_t stream, int xcd_swizzle) {
   if (M % BM != 0 || N % BN != 0 || K % BK != 0)
     throw std::runtime_error(
         "gemm_i8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  const char* var = std::getenv("HPK_GEMM_VARIANT");
+  const bool ph8 = !var || std::string(var) == "8ph";
+  if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 256 == 0) {
+    int tn = (int)(N / 256);
+    int n8 = (int)(M / 256) * tn;
+    hipLaunchKernelGGL(k_gemm_i8_8ph, dim3(n8), dim3(512), 0, stream, C,
+                       (const signed char*)A, (const signed char*)B, (int)M,
+                       (int)N, (int)K, tn, n8, xcd_swizzle);
+    check_hip(hipGetLastError(), "launch_gemm_i8_nt(8ph)");
+    return;
+  }
   int tiles_n = (int)(N / BN);
   int nwg = (int)(M / BM) * tiles_n;
   hipLaunchKernelGGL((k_gemm_i8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,

@@ -116,7 +116,12 @@ def main():
         s1 = torch.full((sz, sz // 32), 127, dtype=torch.uint8, device=dev)
         t = time_gpu(lambda: ops.gemm_mxfp8(c, a8, b8, s1, s1))
         print(f"gemm {sz:5d} mx8  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
-        del a, b, a8, b8, s1, c
+        ai = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
+        bi = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
+        ci = torch.empty(sz, sz, dtype=torch.int32, device=dev)
+        t = time_gpu(lambda: ops.gemm_i8(ci, ai, bi))
+        print(f"gemm {sz:5d} i8   {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TOP/s")
+        del a, b, a8, b8, s1, ai, bi, ci, c
 
 
 if __name__ == "__main__":

@@ -561,3 +561,18 @@ def test_gemm_i8_identity():
     ops.gemm_i8(c, a, b)
     torch.cuda.synchronize()
     assert torch.equal(c, b.int().t())
+
+
+def test_gemm_i8_8phase_exact():
+    """The 8-phase i8 path (256-divisible shapes): full-range exactness."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(61)
+    m, n, k = 256, 512, 512
+    ah = torch.randint(-128, 128, (m, k), generator=g, dtype=torch.int8)
+    bh = torch.randint(-128, 128, (n, k), generator=g, dtype=torch.int8)
+    c = torch.empty(m, n, dtype=torch.int32, device="cuda")
+    ops.gemm_i8(c, ah.cuda(), bh.cuda())
+    ref = torch.matmul(ah.long(), bh.long().t()).to(torch.int32).cuda()
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
