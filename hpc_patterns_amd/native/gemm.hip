@@ -26,6 +26,9 @@
 //                             e8m0 dequant; scale lane layout
 //                             reverse-engineered on hardware —
 //                             scripts/probes/): 1216 TF
+//   k_gemm_i8_nt / _8ph       int8 with EXACT int32 accumulation
+//                             (mfma_i32_16x16x64_i8, ~2x bf16 rate,
+//                             4-VGPR fragments): 1522 / 2214 TOPS
 //
 // Fragment mappings (16x16x32 bf16/fp8): A/B lane L holds 8 contiguous K
 // elements at k = 8*(L>>4), row/col = L&15; C/D lane L reg r holds
