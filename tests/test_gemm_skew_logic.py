@@ -1,0 +1,96 @@
+"""CPU guards for the GEMM LDS-skew bijections (native/gemm.hip).
+
+The skews are pure integer maps; these tests replicate them and assert
+(a) skew/unskew are inverse bijections on the whole image, and (b) the
+bank-window distinctness property each skew was derived for — so a future
+edit that silently breaks the conflict-freedom fails CI without a GPU.
+"""
+
+
+def bf16_skew(e):  # [128][64] bf16 elements
+    row, k = e >> 6, e & 63
+    return (row << 6) | ((k + 16 * ((row >> 1) & 3)) & 63)
+
+
+def bf16_unskew(y):
+    row, k = y >> 6, y & 63
+    return (row << 6) | ((k - 16 * ((row >> 1) & 3)) & 63)
+
+
+def fp8_skew(e):  # [128][64] bytes
+    row, k = e >> 6, e & 63
+    return (row << 6) | ((k + 16 * ((row >> 2) & 3)) & 63)
+
+
+def mx_skew(e):  # [128][128] bytes
+    row, k = e >> 7, e & 127
+    return (row << 7) | ((k + 16 * ((row >> 1) & 5)) & 127)
+
+
+def i8_skew(e):  # [128][64] bytes (plain i8)
+    row, k = e >> 6, e & 63
+    return (row << 6) | ((k + 32 * ((row >> 3) & 1)) & 63)
+
+
+def i8s_skew(e):  # [128][128] bytes (8-phase i8)
+    row, k = e >> 7, e & 127
+    return (row << 7) | ((k + 32 * ((row >> 1) & 3)) & 127)
+
+
+def test_skews_are_row_preserving_bijections():
+    for skew, size, rowshift in ((bf16_skew, 128 * 64, 6),
+                                 (fp8_skew, 128 * 64, 6),
+                                 (mx_skew, 128 * 128, 7),
+                                 (i8_skew, 128 * 64, 6),
+                                 (i8s_skew, 128 * 128, 7)):
+        img = [skew(e) for e in range(size)]
+        assert sorted(img) == list(range(size)), skew.__name__
+        assert all((skew(e) >> rowshift) == (e >> rowshift)
+                   for e in range(size)), skew.__name__
+
+
+def test_bf16_unskew_inverts_skew():
+    for e in range(128 * 64):
+        assert bf16_unskew(bf16_skew(e)) == e
+
+
+def _bank_window(byte_addr, width_dwords=4):
+    return (byte_addr // 4) % 64 // width_dwords
+
+
+def test_bf16_true_lane_groups_conflict_free():
+    """The gfx950 ds_read_b128 lane groups (MI355X_MICROARCH.md): all 16
+    lanes of each group must land on 16 distinct 4-dword windows."""
+    groups = [
+        [0, 1, 2, 3, 12, 13, 14, 15, 20, 21, 22, 23, 24, 25, 26, 27],
+        [4, 5, 6, 7, 8, 9, 10, 11, 16, 17, 18, 19, 28, 29, 30, 31],
+        [32, 33, 34, 35, 44, 45, 46, 47, 52, 53, 54, 55, 56, 57, 58, 59],
+        [36, 37, 38, 39, 40, 41, 42, 43, 48, 49, 50, 51, 60, 61, 62, 63],
+    ]
+    for kk in (0, 32):  # both MFMA K-steps of a 64-element tile row
+        for grp in groups:
+            windows = set()
+            for lane in grp:
+                row = lane & 15
+                kfrag = kk + 8 * (lane >> 4)
+                e = bf16_skew(row * 64 + kfrag)  # element index
+                windows.add(_bank_window((e & 63) * 2 + (e >> 6) * 128))
+            assert len(windows) == 16, (kk, grp, sorted(windows))
+
+
+def test_i8s_lane_groups_conflict_free():
+    groups = [
+        [0, 1, 2, 3, 12, 13, 14, 15, 20, 21, 22, 23, 24, 25, 26, 27],
+        [4, 5, 6, 7, 8, 9, 10, 11, 16, 17, 18, 19, 28, 29, 30, 31],
+        [32, 33, 34, 35, 44, 45, 46, 47, 52, 53, 54, 55, 56, 57, 58, 59],
+        [36, 37, 38, 39, 40, 41, 42, 43, 48, 49, 50, 51, 60, 61, 62, 63],
+    ]
+    for kk in (0, 64):
+        for grp in groups:
+            windows = set()
+            for lane in grp:
+                row = lane & 15
+                kfrag = kk + 16 * (lane >> 4)
+                y = i8s_skew(row * 128 + kfrag)
+                windows.add(_bank_window((y & 127) + (y >> 7) * 128))
+            assert len(windows) == 16, (kk, grp, sorted(windows))
