@@ -83,9 +83,9 @@ __device__ __forceinline__ long lds_unskew(long y) { // LDS slot -> tile elem
 // Wave-grid decomposition is a template knob: <2,2> = 4 waves of 64x64
 // (4x4 fragments, 190 VGPR+AGPR, 2 waves/SIMD), <2,4> = 8 waves of
 // 64x32 (4x2 fragments, 76 VGPR, 6 waves/SIMD). Measured on MI355X with
-// random operands: 8 waves wins everywhere (924 vs 765 TF at 8192^3) —
-// occupancy-driven latency hiding beats the bigger per-wave MFMA batch.
-// HPK_GEMM_WAVES=4|8 overrides for experiments.
+// random operands: 8 waves wins everywhere (935 vs 765 TF at 8192^3
+// under the final skew) — occupancy-driven latency hiding beats the
+// bigger per-wave MFMA batch. HPK_GEMM_WAVES=4|8 overrides.
 template <int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
@@ -129,7 +129,6 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     __syncthreads(); // previous K-step's reads done before overwrite
     for (int issue = 0; issue < ISSUES; ++issue) {
       long o_base = (long)issue * elems_per_issue + (long)wid * (64 * 8);
-      (void)0;
       // lane's LDS slot is o_base + lane*8 (lane-linear); fetch the global
       // chunk that belongs at that slot under the skewed image
       long o = lds_unskew(o_base + (long)lane * 8);
