@@ -112,12 +112,14 @@ def gemm_bf16(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
     """K7 (r2): C[M,N] fp32 = A[M,K] @ B[N,K]^T, A/B bf16 K-contiguous.
 
     Hand-written LDS-tiled v_mfma_f32_16x16x32_bf16 kernel (native/gemm.hip):
-    128x128 tile per 8-wave workgroup (64x32 per wave), cyclic-skew LDS
-    layout (conflict-free ds_read_b128), 16-byte global_load_lds staging.
-    Measured 924 TF bf16 at 8192^3 on random operands (37% of the 2.5 PF
-    dense peak; profiles/gemm_showcase_r2.log). xcd_swizzle enables the
-    bijective XCD workgroup remap (measured slower on the default variant,
-    off by default). Requires M,N multiples of 128 and K a multiple of 64.
+    zero-bank-conflict cyclic-skew LDS layout, 16-byte global_load_lds
+    staging, and (for M,N % 256, K % 128 shapes) the 256^2-tile 8-phase
+    counted-vmcnt pipeline selected by default — 1085 TF bf16 at 8192^3 on
+    random operands (43% of the 2.5 PF dense peak;
+    profiles/gemm_showcase_r2.log). HPK_GEMM_VARIANT=plain|db|8ph forces a
+    variant. xcd_swizzle enables the bijective XCD workgroup remap
+    (measured slower here, off by default). Requires M,N multiples of 128
+    and K a multiple of 64 (use matmul_nt for arbitrary shapes).
     """
     if c.dtype != torch.float32 or a.dtype != torch.bfloat16 \
             or b.dtype != torch.bfloat16:
@@ -209,3 +211,79 @@ def gemm_i8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
                          f" C{tuple(c.shape)}")
     native().gemm_i8_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
                         m, n, k, _stream_handle(stream), int(xcd_swizzle))
+
+
+def gemm_pad_shapes(kind: str, m: int, n: int, k: int):
+    """Padded (M, N, K) that puts a (m, n, k) problem on the fast path.
+
+    kind is "bf16" / "fp8" (8-phase needs M,N % 256, K % 128), "i8"
+    (M,N % 256, K % 256) or "mxfp8" (plain kernel only: M,N,K % 128).
+    Pure shape math — unit-tested on CPU (tests/test_gemm_skew_logic.py).
+    """
+    if kind not in ("bf16", "fp8", "i8", "mxfp8"):
+        raise ValueError(f"unknown gemm kind {kind!r}")
+    def up(x, q):
+        return -(-x // q) * q
+    if kind == "mxfp8":
+        return up(m, 128), up(n, 128), up(k, 128)
+    kq = 256 if kind == "i8" else 128
+    return up(m, 256), up(n, 256), up(k, kq)
+
+
+def _pad2d(t: torch.Tensor, rows: int, cols: int, fill=0):
+    if t.shape == (rows, cols):
+        return t
+    out = torch.zeros(rows, cols, dtype=t.dtype, device=t.device) if fill == 0 \
+        else torch.full((rows, cols), fill, dtype=t.dtype, device=t.device)
+    out[: t.shape[0], : t.shape[1]] = t
+    return out
+
+
+def matmul_nt(a: torch.Tensor, b: torch.Tensor,
+              a_scale: torch.Tensor = None, b_scale: torch.Tensor = None,
+              stream=None, xcd_swizzle: bool = False) -> torch.Tensor:
+    """Arbitrary-shape front door to the K7 GEMM family: returns A @ B^T.
+
+    Dispatches on dtype — bf16 -> gemm_bf16, float8_e4m3fn -> gemm_fp8
+    (or gemm_mxfp8 when e8m0 scales are given), int8 -> gemm_i8 — after
+    zero-padding the operands up to the fast-path tile multiples
+    (gemm_pad_shapes). Zero rows/columns contribute nothing, so the
+    result is bit-identical to the unpadded kernel output; the padded
+    region of C is sliced away (result is a fresh contiguous tensor when
+    padding occurred). Scale padding uses 127 (= 2^0) — irrelevant since
+    the padded data is zero.
+    """
+    if a.dim() != 2 or b.dim() != 2 or a.shape[1] != b.shape[1]:
+        raise ValueError(f"need A[M,K], B[N,K]; got A{tuple(a.shape)} "
+                         f"B{tuple(b.shape)}")
+    m, k = a.shape
+    n = b.shape[0]
+    if a.dtype == torch.bfloat16:
+        kind, fn, out_dtype = "bf16", gemm_bf16, torch.float32
+    elif a.dtype == torch.int8:
+        kind, fn, out_dtype = "i8", gemm_i8, torch.int32
+    elif a.dtype == torch.float8_e4m3fn:
+        if a_scale is not None:
+            kind, fn, out_dtype = "mxfp8", None, torch.float32
+        else:
+            kind, fn, out_dtype = "fp8", gemm_fp8, torch.float32
+    else:
+        raise TypeError(f"unsupported operand dtype {a.dtype}")
+    mp, np_, kp = gemm_pad_shapes(kind, m, n, k)
+    ap = _pad2d(a.contiguous(), mp, kp)
+    bp = _pad2d(b.contiguous(), np_, kp)
+    c = torch.empty(mp, np_, dtype=out_dtype, device=a.device)
+    if kind == "mxfp8":
+        if b_scale is None or a_scale.shape != (m, k // 32) \
+                or b_scale.shape != (n, k // 32):
+            raise ValueError("mx path needs a_scale [M,K//32] and "
+                             "b_scale [N,K//32] (K % 32 == 0)")
+        asp = _pad2d(a_scale.contiguous(), mp, kp // 32, fill=127)
+        bsp = _pad2d(b_scale.contiguous(), np_, kp // 32, fill=127)
+        gemm_mxfp8(c, ap, bp, asp, bsp, stream=stream,
+                   xcd_swizzle=xcd_swizzle)
+    else:
+        fn(c, ap, bp, stream=stream, xcd_swizzle=xcd_swizzle)
+    if (mp, np_) == (m, n):
+        return c
+    return c[:m, :n].contiguous()

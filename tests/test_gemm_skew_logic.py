@@ -94,3 +94,33 @@ def test_i8s_lane_groups_conflict_free():
                 y = i8s_skew(row * 128 + kfrag)
                 windows.add(_bank_window((y & 127) + (y >> 7) * 128))
             assert len(windows) == 16, (kk, grp, sorted(windows))
+
+
+# ---------------------------------------------------------------------------
+# matmul_nt padding math (ops.gemm_pad_shapes) — pure CPU
+# ---------------------------------------------------------------------------
+
+def test_gemm_pad_shapes():
+    from hpc_patterns_amd.ops import gemm_pad_shapes
+
+    # already-aligned shapes are untouched
+    assert gemm_pad_shapes("bf16", 256, 512, 128) == (256, 512, 128)
+    assert gemm_pad_shapes("i8", 256, 256, 256) == (256, 256, 256)
+    assert gemm_pad_shapes("mxfp8", 128, 128, 128) == (128, 128, 128)
+    # odd shapes round UP to the fast-path multiples
+    assert gemm_pad_shapes("bf16", 300, 520, 130) == (512, 768, 256)
+    assert gemm_pad_shapes("fp8", 1, 1, 1) == (256, 256, 128)
+    assert gemm_pad_shapes("i8", 100, 300, 700) == (256, 512, 768)
+    assert gemm_pad_shapes("mxfp8", 129, 127, 257) == (256, 128, 384)
+    # padded result always satisfies the kernel's divisibility contract
+    for kind, (qm, qk) in (("bf16", (256, 128)), ("fp8", (256, 128)),
+                           ("i8", (256, 256)), ("mxfp8", (128, 128))):
+        for m, n, k in ((1, 1, 1), (300, 520, 736), (511, 513, 257)):
+            mp, np_, kp = gemm_pad_shapes(kind, m, n, k)
+            assert mp % (128 if kind == "mxfp8" else qm) == 0
+            assert np_ % (128 if kind == "mxfp8" else qm) == 0
+            assert kp % qk == 0
+            assert mp >= m and np_ >= n and kp >= k
+    import pytest
+    with pytest.raises(ValueError):
+        gemm_pad_shapes("fp16", 128, 128, 128)

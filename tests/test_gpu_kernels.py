@@ -576,3 +576,74 @@ def test_gemm_i8_8phase_exact():
     ref = torch.matmul(ah.long(), bh.long().t()).to(torch.int32).cuda()
     torch.cuda.synchronize()
     assert torch.equal(c, ref)
+
+
+# ---------------------------------------------------------------------------
+# matmul_nt — arbitrary-shape front door (zero-padding dispatch)
+# ---------------------------------------------------------------------------
+
+def test_matmul_nt_bf16_odd_shapes_exact():
+    """Odd (non-tile-multiple) shapes through the padding path: small
+    integers keep everything exact, so the sliced result must EQUAL the
+    torch fp32 reference."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(71)
+    for m, n, k in ((300, 520, 736), (1, 1, 64), (257, 129, 192)):
+        a = torch.randint(-4, 5, (m, k), generator=g).to(torch.bfloat16).cuda()
+        b = torch.randint(-4, 5, (n, k), generator=g).to(torch.bfloat16).cuda()
+        c = ops.matmul_nt(a, b)
+        ref = torch.matmul(a.float(), b.float().t())
+        torch.cuda.synchronize()
+        assert c.shape == (m, n) and c.dtype == torch.float32
+        assert torch.equal(c, ref), (m, n, k)
+
+
+def test_matmul_nt_i8_odd_shapes_exact():
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(73)
+    m, n, k = 100, 300, 700
+    ah = torch.randint(-128, 128, (m, k), generator=g, dtype=torch.int8)
+    bh = torch.randint(-128, 128, (n, k), generator=g, dtype=torch.int8)
+    c = ops.matmul_nt(ah.cuda(), bh.cuda())
+    ref = torch.matmul(ah.long(), bh.long().t()).to(torch.int32).cuda()
+    torch.cuda.synchronize()
+    assert c.shape == (m, n) and c.dtype == torch.int32
+    assert torch.equal(c, ref)
+
+
+def test_matmul_nt_fp8_and_mx_odd_shapes():
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(79)
+    m, n, k = 200, 136, 160
+    a = torch.randint(-4, 5, (m, k), generator=g).float() \
+        .to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).float() \
+        .to(torch.float8_e4m3fn).cuda()
+    ref = torch.matmul(a.float(), b.float().t())
+    c = ops.matmul_nt(a, b)
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
+    # mx path: unit scales (127 = 2^0) must reproduce the plain result
+    s_a = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    s_b = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    cmx = ops.matmul_nt(a, b, a_scale=s_a, b_scale=s_b)
+    torch.cuda.synchronize()
+    assert torch.equal(cmx, ref)
+
+
+def test_matmul_nt_aligned_no_copy():
+    """Tile-multiple shapes skip padding entirely: result tensor IS the
+    kernel's output buffer (no slice copy)."""
+    from hpc_patterns_amd import ops
+
+    torch.manual_seed(83)
+    a = torch.randn(256, 128, device="cuda").to(torch.bfloat16)
+    b = torch.randn(256, 128, device="cuda").to(torch.bfloat16)
+    c = ops.matmul_nt(a, b)
+    torch.cuda.synchronize()
+    ref = torch.matmul(a.float(), b.float().t())
+    assert c.is_contiguous() and c.shape == (256, 256)
+    assert torch.allclose(c, ref, rtol=1e-3, atol=1e-3)
