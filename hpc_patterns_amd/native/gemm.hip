@@ -80,6 +80,28 @@ __device__ __forceinline__ long lds_unskew(long y) { // LDS slot -> tile elem
   return (row << 6) | ((k - 16 * ((row >> 1) & 3)) & 63);
 }
 
+// L2/MALL-aware grouped tile order (the CUTLASS threadblock-swizzle idea,
+// re-derived for the 256 MiB MALL): remap the linear workgroup id so the
+// ~256 CONCURRENTLY-resident workgroups cover a near-square super-block of
+// output tiles — their A row-panels and B column-panels then stay resident
+// in the LLC and each panel is streamed from HBM once per super-block
+// instead of once per tile row. `group` = band width in N-tiles
+// (HPK_GEMM_GROUP; <=1 keeps the row-major order). Bijective for any
+// grid: the last band is simply narrower (gw < group).
+__device__ __forceinline__ int hpk_group_remap(int wg, int tiles_n, int nwg,
+                                               int group) {
+  if (group <= 1) return wg;
+  const int tiles_m = nwg / tiles_n;
+  const int band = group * tiles_m;
+  const int b = wg / band;
+  const int within = wg - b * band;
+  int gw = tiles_n - b * group;
+  if (gw > group) gw = group;
+  const int tm = within / gw;
+  const int tn = b * group + within % gw;
+  return tm * tiles_n + tn;
+}
+
 // Wave-grid decomposition is a template knob: <2,2> = 4 waves of 64x64
 // (4x4 fragments, 190 VGPR+AGPR, 2 waves/SIMD), <2,4> = 8 waves of
 // 64x32 (4x2 fragments, 76 VGPR, 6 waves/SIMD). Measured on MI355X with
@@ -90,7 +112,7 @@ template <int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
     const __hip_bfloat16* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   constexpr int THREADS = WAVES_M * WAVES_N * 64;
   constexpr int MREP = BM / (WAVES_M * 16);
   constexpr int NREP = BN / (WAVES_N * 16);
@@ -105,6 +127,7 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const int tile_m = wg / tiles_n;
   const int tile_n = wg % tiles_n;
   const long brow = (long)tile_m * BM;
@@ -188,7 +211,7 @@ template <int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt_db(
     float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
     const __hip_bfloat16* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   constexpr int THREADS = WAVES_M * WAVES_N * 64;
   constexpr int MREP = BM / (WAVES_M * 16);
   constexpr int NREP = BN / (WAVES_N * 16);
@@ -202,6 +225,7 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_bf16_nt_db(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * BM;
   const long bcol = (long)(wg % tiles_n) * BN;
   const int tid = threadIdx.x;
@@ -310,7 +334,7 @@ template <int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_fp8_nt(
     float* __restrict__ C, const unsigned char* __restrict__ A,
     const unsigned char* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   constexpr int THREADS = WAVES_M * WAVES_N * 64;
   constexpr int MREP = BM / (WAVES_M * 16);
   constexpr int NREP = BN / (WAVES_N * 16);
@@ -324,6 +348,7 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_fp8_nt(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * BM;
   const long bcol = (long)(wg % tiles_n) * BN;
   const int tid = threadIdx.x;
@@ -423,7 +448,7 @@ constexpr int PHALF = 128 * 64; // elements per half-tile image (16 KiB)
 __global__ __launch_bounds__(512) void k_gemm_bf16_8ph(
     float* __restrict__ C, const __hip_bfloat16* __restrict__ A,
     const __hip_bfloat16* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   __shared__ __hip_bfloat16 lds[8 * PHALF]; // 128 KiB
 
   int wg = (int)blockIdx.x;
@@ -432,6 +457,7 @@ __global__ __launch_bounds__(512) void k_gemm_bf16_8ph(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * 256;
   const long bcol = (long)(wg % tiles_n) * 256;
   const int tid = threadIdx.x;
@@ -604,7 +630,7 @@ constexpr int PHALF8 = 128 * 64; // fp8 half-tile image (8 KiB)
 __global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
     float* __restrict__ C, const unsigned char* __restrict__ A,
     const unsigned char* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   __shared__ unsigned char lds[8 * PHALF8]; // 64 KiB -> 2 blocks/CU
 
   int wg = (int)blockIdx.x;
@@ -613,6 +639,7 @@ __global__ __launch_bounds__(512) void k_gemm_fp8_8ph(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * 256;
   const long bcol = (long)(wg % tiles_n) * 256;
   const int tid = threadIdx.x;
@@ -814,7 +841,7 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt(
     float* __restrict__ C, const unsigned char* __restrict__ A,
     const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
     const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
-    int nwg, int xcd_swizzle) {
+    int nwg, int xcd_swizzle, int group) {
   constexpr int MREP = 4, NREP = 2; // 8 waves as 2x4, 64x32 per wave
   // [A data 16K][B data 16K][A scales 512][B scales 512]
   __shared__ unsigned char lds[2 * 128 * MXK + 2 * 512];
@@ -825,6 +852,7 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * 128;
   const long bcol = (long)(wg % tiles_n) * 128;
   const int tid = threadIdx.x;
@@ -945,7 +973,7 @@ template <int WAVES_M, int WAVES_N>
 __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_i8_nt(
     int* __restrict__ C, const signed char* __restrict__ A,
     const signed char* __restrict__ B, int M, int N, int K, int tiles_n,
-    int nwg, int xcd_swizzle) {
+    int nwg, int xcd_swizzle, int group) {
   constexpr int THREADS = WAVES_M * WAVES_N * 64;
   constexpr int MREP = BM / (WAVES_M * 16);
   constexpr int NREP = BN / (WAVES_N * 16);
@@ -959,6 +987,7 @@ __global__ __launch_bounds__(WAVES_M* WAVES_N * 64) void k_gemm_i8_nt(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * BM;
   const long bcol = (long)(wg % tiles_n) * BN;
   const int tid = threadIdx.x;
@@ -1050,7 +1079,7 @@ constexpr int PH8I = 128 * 128; // i8 half-tile image bytes (16 KiB)
 __global__ __launch_bounds__(512) void k_gemm_i8_8ph(
     int* __restrict__ C, const signed char* __restrict__ A,
     const signed char* __restrict__ B, int M, int N, int K,
-    int tiles_n, int nwg, int xcd_swizzle) {
+    int tiles_n, int nwg, int xcd_swizzle, int group) {
   __shared__ signed char lds[8 * PH8I]; // 128 KiB
 
   int wg = (int)blockIdx.x;
@@ -1059,6 +1088,7 @@ __global__ __launch_bounds__(512) void k_gemm_i8_8ph(
     int xcd = wg % 8, i = wg / 8;
     wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
   }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
   const long brow = (long)(wg / tiles_n) * 256;
   const long bcol = (long)(wg % tiles_n) * 256;
   const int tid = threadIdx.x;
@@ -1222,11 +1252,23 @@ __global__ __launch_bounds__(512) void k_gemm_i8_8ph(
     }
 }
 
+// Band width (in N-tiles) of the grouped tile order above. Defaults come
+// from interleaved A/B sweeps at 8192^3 and 16384^3 (clock-drift-controlled;
+// docs/gemm.md): row-major is already LLC-friendly up to ~32 N-tiles, wider
+// grids gain 4-7% from 16/32-wide bands. HPK_GEMM_GROUP overrides.
+static int gemm_group(int tiles_n) {
+  if (const char* env = std::getenv("HPK_GEMM_GROUP")) return std::atoi(env);
+  if (tiles_n >= 128) return 32;
+  if (tiles_n >= 64) return 16;
+  return 1;
+}
+
 void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
                        long N, long K, hipStream_t stream, int xcd_swizzle) {
   if (M % BM != 0 || N % BN != 0 || K % BK != 0)
     throw std::runtime_error(
         "gemm_i8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  const int grp = gemm_group((int)(N / 256));
   const char* var = std::getenv("HPK_GEMM_VARIANT");
   const bool ph8 = !var || std::string(var) == "8ph";
   if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 256 == 0) {
@@ -1234,7 +1276,7 @@ void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
     int n8 = (int)(M / 256) * tn;
     hipLaunchKernelGGL(k_gemm_i8_8ph, dim3(n8), dim3(512), 0, stream, C,
                        (const signed char*)A, (const signed char*)B, (int)M,
-                       (int)N, (int)K, tn, n8, xcd_swizzle);
+                       (int)N, (int)K, tn, n8, xcd_swizzle, grp);
     check_hip(hipGetLastError(), "launch_gemm_i8_nt(8ph)");
     return;
   }
@@ -1242,7 +1284,7 @@ void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
   int nwg = (int)(M / BM) * tiles_n;
   hipLaunchKernelGGL((k_gemm_i8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,
                      C, (const signed char*)A, (const signed char*)B, (int)M,
-                     (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+                     (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
   check_hip(hipGetLastError(), "launch_gemm_i8_nt");
 }
 
@@ -1251,12 +1293,13 @@ void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                           long K, hipStream_t stream, int xcd_swizzle) {
   if (M % 128 != 0 || N % 128 != 0 || K % 128 != 0)
     throw std::runtime_error("gemm_mxfp8_nt requires M,N,K % 128 == 0");
+  const int grp = gemm_group((int)(N / 128));
   int tiles_n = (int)(N / 128);
   int nwg = (int)(M / 128) * tiles_n;
   hipLaunchKernelGGL(k_gemm_mxfp8_nt, dim3(nwg), dim3(512), 0, stream, C,
                      (const unsigned char*)A, (const unsigned char*)B,
                      (const unsigned char*)As, (const unsigned char*)Bs,
-                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
   check_hip(hipGetLastError(), "launch_gemm_mxfp8_nt");
 }
 
@@ -1266,6 +1309,7 @@ void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
   if (M % BM != 0 || N % BN != 0 || K % BK != 0)
     throw std::runtime_error(
         "gemm_fp8_nt requires M,N % 128 == 0 and K % 64 == 0");
+  const int grp = gemm_group((int)(N / 256));
   const char* var = std::getenv("HPK_GEMM_VARIANT");
   const bool ph8 = !var || std::string(var) == "8ph";
   if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0) {
@@ -1273,7 +1317,7 @@ void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
     int n8 = (int)(M / 256) * tn;
     hipLaunchKernelGGL(k_gemm_fp8_8ph, dim3(n8), dim3(512), 0, stream, C,
                        (const unsigned char*)A, (const unsigned char*)B,
-                       (int)M, (int)N, (int)K, tn, n8, xcd_swizzle);
+                       (int)M, (int)N, (int)K, tn, n8, xcd_swizzle, grp);
     check_hip(hipGetLastError(), "launch_gemm_fp8_nt(8ph)");
     return;
   }
@@ -1281,7 +1325,7 @@ void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
   int nwg = tiles_m * tiles_n;
   hipLaunchKernelGGL((k_gemm_fp8_nt<2, 4>), dim3(nwg), dim3(512), 0, stream,
                      C, (const unsigned char*)A, (const unsigned char*)B,
-                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle);
+                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
   check_hip(hipGetLastError(), "launch_gemm_fp8_nt");
 }
 
@@ -1291,6 +1335,7 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
   if (M % BM != 0 || N % BN != 0 || K % BK != 0)
     throw std::runtime_error(
         "gemm_bf16_nt requires M,N % 128 == 0 and K % 64 == 0");
+  const int grp = gemm_group((int)(N / 256));
   int tiles_m = (int)(M / BM), tiles_n = (int)(N / BN);
   int nwg = tiles_m * tiles_n;
   // measured (profiles/gemm_r2 logs, random [-1,1) operands): 8 waves
@@ -1311,7 +1356,7 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
     int n8 = (int)(M / 256) * tn;
     hipLaunchKernelGGL(k_gemm_bf16_8ph, dim3(n8), dim3(512), 0, stream, C,
                        (const __hip_bfloat16*)A, (const __hip_bfloat16*)B,
-                       (int)M, (int)N, (int)K, tn, n8, xcd_swizzle);
+                       (int)M, (int)N, (int)K, tn, n8, xcd_swizzle, grp);
     check_hip(hipGetLastError(), "launch_gemm_bf16_nt(8ph)");
     return;
   }
@@ -1319,22 +1364,22 @@ void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
     hipLaunchKernelGGL((k_gemm_bf16_nt_db<2, 4>), dim3(nwg), dim3(512), 0,
                        stream, C, (const __hip_bfloat16*)A,
                        (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
-                       tiles_n, nwg, xcd_swizzle);
+                       tiles_n, nwg, xcd_swizzle, grp);
   } else if (dbuf) {
     hipLaunchKernelGGL((k_gemm_bf16_nt_db<2, 2>), dim3(nwg), dim3(256), 0,
                        stream, C, (const __hip_bfloat16*)A,
                        (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
-                       tiles_n, nwg, xcd_swizzle);
+                       tiles_n, nwg, xcd_swizzle, grp);
   } else if (waves == 8) {
     hipLaunchKernelGGL((k_gemm_bf16_nt<2, 4>), dim3(nwg), dim3(512), 0,
                        stream, C, (const __hip_bfloat16*)A,
                        (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
-                       tiles_n, nwg, xcd_swizzle);
+                       tiles_n, nwg, xcd_swizzle, grp);
   } else {
     hipLaunchKernelGGL((k_gemm_bf16_nt<2, 2>), dim3(nwg), dim3(256), 0,
                        stream, C, (const __hip_bfloat16*)A,
                        (const __hip_bfloat16*)B, (int)M, (int)N, (int)K,
-                       tiles_n, nwg, xcd_swizzle);
+                       tiles_n, nwg, xcd_swizzle, grp);
   }
   check_hip(hipGetLastError(), "launch_gemm_bf16_nt");
 }

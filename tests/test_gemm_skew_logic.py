@@ -124,3 +124,29 @@ def test_gemm_pad_shapes():
     import pytest
     with pytest.raises(ValueError):
         gemm_pad_shapes("fp16", 128, 128, 128)
+
+
+def test_group_remap_bijective():
+    """Replicates hpk_group_remap (gemm.hip): the grouped tile order must
+    be a bijection on [0, nwg) for every band width, including grids whose
+    tiles_n is not a multiple of the group."""
+    def remap(wg, tiles_n, nwg, group):
+        if group <= 1:
+            return wg
+        tiles_m = nwg // tiles_n
+        band = group * tiles_m
+        b = wg // band
+        within = wg - b * band
+        gw = min(group, tiles_n - b * group)
+        return (within // gw) * tiles_n + b * group + within % gw
+
+    for tiles_m, tiles_n in ((32, 32), (4, 7), (7, 4), (1, 5), (5, 1),
+                             (16, 48), (3, 3)):
+        nwg = tiles_m * tiles_n
+        for group in (1, 2, 3, 4, 8, 16, 64):
+            out = {remap(w, tiles_n, nwg, group) for w in range(nwg)}
+            assert out == set(range(nwg)), (tiles_m, tiles_n, group)
+    # grouped order covers a band of `group` columns before moving on:
+    # the first tiles_m remapped ids all live in columns [0, group)
+    first = {remap(w, 32, 32 * 32, 8) % 32 for w in range(8 * 32)}
+    assert first == set(range(8))

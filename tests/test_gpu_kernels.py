@@ -647,3 +647,42 @@ def test_matmul_nt_aligned_no_copy():
     ref = torch.matmul(a.float(), b.float().t())
     assert c.is_contiguous() and c.shape == (256, 256)
     assert torch.allclose(c, ref, rtol=1e-3, atol=1e-3)
+
+
+def test_gemm_grouped_order_exact(monkeypatch):
+    """The L2-aware grouped tile order (HPK_GEMM_GROUP) is a pure launch
+    reordering: results must stay bitwise identical, including grids whose
+    tiles_n is not a multiple of the band width (partial last band)."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(89)
+    m, n, k = 768, 1792, 512  # 8ph grid: 3 x 7 tiles
+    a = torch.randint(-4, 5, (m, k), generator=g).to(torch.bfloat16).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).to(torch.bfloat16).cuda()
+    ref = torch.matmul(a.float(), b.float().t())
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    for grp in ("2", "4", "8", "16", "64"):
+        monkeypatch.setenv("HPK_GEMM_GROUP", grp)
+        ops.gemm_bf16(c, a, b)
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), grp
+
+
+def test_gemm_mxfp8_grouped_order_exact(monkeypatch):
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(97)
+    m, n, k = 384, 1152, 256  # 128-tile grid: 3 x 9
+    a = torch.randint(-4, 5, (m, k), generator=g).float() \
+        .to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-4, 5, (n, k), generator=g).float() \
+        .to(torch.float8_e4m3fn).cuda()
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    ref = torch.matmul(a.float(), b.float().t())
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    for grp in ("4", "16", "32"):
+        monkeypatch.setenv("HPK_GEMM_GROUP", grp)
+        ops.gemm_mxfp8(c, a, b, sa, sb)
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), grp
