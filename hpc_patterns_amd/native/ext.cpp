@@ -99,6 +99,19 @@ PYBIND11_MODULE(_hpk, m) {
         py::arg("c"), py::arg("a"), py::arg("b"), py::arg("a_scale"),
         py::arg("b_scale"), py::arg("m"), py::arg("n"), py::arg("k"),
         py::arg("stream") = 0, py::arg("xcd_swizzle") = 0);
+  m.def("gemm_mxfp4_nt",
+        [](uintptr_t c, uintptr_t a, uintptr_t b, uintptr_t as, uintptr_t bs,
+           long m, long n, long k, uintptr_t stream, int xcd_swizzle) {
+          hpk::launch_gemm_mxfp4_nt(reinterpret_cast<float*>(c),
+                                    reinterpret_cast<const void*>(a),
+                                    reinterpret_cast<const void*>(b),
+                                    reinterpret_cast<const void*>(as),
+                                    reinterpret_cast<const void*>(bs), m, n,
+                                    k, as_stream(stream), xcd_swizzle);
+        },
+        py::arg("c"), py::arg("a"), py::arg("b"), py::arg("a_scale"),
+        py::arg("b_scale"), py::arg("m"), py::arg("n"), py::arg("k"),
+        py::arg("stream") = 0, py::arg("xcd_swizzle") = 0);
   m.def("copy_kernel",
         [](uintptr_t dst, uintptr_t src, size_t nbytes, uintptr_t stream) {
           hpk::launch_copy_kernel(reinterpret_cast<void*>(dst),

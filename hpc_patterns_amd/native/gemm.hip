@@ -1288,6 +1288,112 @@ void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
   check_hip(hipGetLastError(), "launch_gemm_i8_nt");
 }
 
+
+// ---------------------------------------------------------------------------
+// K7-mxfp4: block-scaled OCP MX-fp4 GEMM — the same scaled MFMA in fp4
+// (e2m1) mode, the 4x-bf16 rate class (~10 PF dense spec; fp4 shares the
+// fp6 rate on CDNA4). fp4_probe2/fp4_probe1 (scripts/probes/) measured a
+// DIAGONAL operand/scale layout — unlike fp8's cross-lane pattern: data
+// lane (row, g) supplies the ONE contiguous OCP 32-block k in
+// [32g, 32g+32) packed 2 elements/byte (within-block nibble order is a
+// free consistent K-permutation — both orders validated), and the lane's
+// OWN scale operand byte 0 (opsel 0) covers exactly that block. Each
+// fragment is therefore ONE contiguous 16-byte LDS read, and the natural
+// [128 rows][64 bytes] tile is bank-conflict-free for the true b128 lane
+// groups (window (4*row + g) mod 32 — enumerated in
+// tests/test_gemm_skew_logic.py): NO skew is needed. Operands live in
+// the low 4 VGPRs of the v8i32 builtin argument (fp4 reads 16 B/lane).
+constexpr int MX4B = 64; // packed bytes per row per K-tile (128 elements)
+__global__ __launch_bounds__(512) void k_gemm_mxfp4_nt(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int MREP = 4, NREP = 2; // 8 waves as 2x4, 64x32 per wave
+  // [A data 8K][B data 8K][A scales 512][B scales 512]
+  __shared__ unsigned char lds[2 * 128 * MX4B + 2 * 512];
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 128;
+  const long bcol = (long)(wg % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+  const int ks = K / 32;       // scale row stride
+  const long Kb = (long)K / 2; // packed data row stride (bytes)
+
+  unsigned char* sA = lds + 2 * 128 * MX4B;
+  unsigned char* sB = sA + 512;
+  f32x4 acc[MREP][NREP] = {};
+
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    __syncthreads();
+    // data: [128][64] bytes per operand = 512 threads x 16 B, one issue
+    {
+      long o_base = (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 6);
+      int kk = (int)(o & 63);
+      const unsigned char* ga = A + (brow + row) * Kb + k0 / 2 + kk;
+      const unsigned char* gb = B + (bcol + row) * Kb + k0 / 2 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(lds + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(lds + 128 * MX4B + o_base),
+          16, 0, 0);
+    }
+    // scales: 128 rows x 4 k-blocks per operand (as in the mx8 kernel)
+    {
+      int row = tid >> 2, kb = tid & 3;
+      sA[tid] = As[(brow + row) * (long)ks + k0 / 32 + kb];
+      sB[tid] = Bs[(bcol + row) * (long)ks + k0 / 32 + kb];
+    }
+    __syncthreads();
+
+    const int g = lane >> 4;
+    typedef __attribute__((ext_vector_type(4))) int i32x4;
+    auto frag16 = [&](const unsigned char* base, long byteoff) {
+      i32x4 lo = *(const i32x4*)__builtin_assume_aligned(base + byteoff, 16);
+      i32x8 f = {};
+      for (int j = 0; j < 4; ++j) f[j] = lo[j];
+      return f;
+    };
+    i32x8 afrag[MREP];
+    int asc[MREP];
+    for (int m = 0; m < MREP; ++m) {
+      int row = wr * 64 + m * 16 + (lane & 15);
+      afrag[m] = frag16(lds, (long)row * MX4B + 16 * g);
+      asc[m] = sA[row * 4 + g];
+    }
+    for (int n = 0; n < NREP; ++n) {
+      int col = wc * 32 + n * 16 + (lane & 15);
+      i32x8 bfrag = frag16(lds + 128 * MX4B, (long)col * MX4B + 16 * g);
+      int bsc = sB[col * 4 + g];
+      for (int m = 0; m < MREP; ++m)
+        acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            afrag[m], bfrag, acc[m][n], 4, 4, 0, asc[m], 0, bsc);
+    }
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * 64 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 32 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle) {
@@ -1301,6 +1407,22 @@ void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                      (const unsigned char*)As, (const unsigned char*)Bs,
                      (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
   check_hip(hipGetLastError(), "launch_gemm_mxfp8_nt");
+}
+
+
+void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
+                          const void* As, const void* Bs, long M, long N,
+                          long K, hipStream_t stream, int xcd_swizzle) {
+  if (M % 128 != 0 || N % 128 != 0 || K % 128 != 0)
+    throw std::runtime_error("gemm_mxfp4_nt requires M,N,K % 128 == 0");
+  const int grp = gemm_group((int)(N / 128));
+  int tiles_n = (int)(N / 128);
+  int nwg = (int)(M / 128) * tiles_n;
+  hipLaunchKernelGGL(k_gemm_mxfp4_nt, dim3(nwg), dim3(512), 0, stream, C,
+                     (const unsigned char*)A, (const unsigned char*)B,
+                     (const unsigned char*)As, (const unsigned char*)Bs,
+                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
+  check_hip(hipGetLastError(), "launch_gemm_mxfp4_nt");
 }
 
 void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,

@@ -89,6 +89,13 @@ void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
 void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle = 0);
+// Block-scaled MX-fp4 (e2m1, the 4x-bf16 rate class): A/B are
+// nibble-PACKED uint8 [rows][K/2] (low nibble = even k), scales as for
+// mxfp8. Operand/scale layout measured on hardware: diagonal (one OCP
+// 32-block per lane, own-lane scale — scripts/probes/fp4_probe*).
+void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
+                          const void* As, const void* Bs, long M, long N,
+                          long K, hipStream_t stream, int xcd_swizzle = 0);
 
 // Exact double-precision sum of n floats. Synchronizes `stream`.
 // Replaces the reference's O(N log N) host sort+sum checksum

@@ -191,6 +191,77 @@ def gemm_mxfp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
                            m, n, k, _stream_handle(stream), int(xcd_swizzle))
 
 
+# the 16 OCP e2m1 (fp4) values, indexed by nibble
+_E2M1_VALUES = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+                -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0]
+
+
+def e2m1_pack(t: torch.Tensor) -> torch.Tensor:
+    """Pack a float tensor of exactly-representable e2m1 values
+    (0, ±0.5, ±1, ±1.5, ±2, ±3, ±4, ±6) into uint8 nibbles, two per byte
+    (low nibble = even index along the last dim). Raises on values
+    outside the e2m1 set — this is a test/packing utility, not a
+    quantizer. Last dim must be even."""
+    if t.shape[-1] % 2 != 0:
+        raise ValueError("last dim must be even")
+    table = torch.tensor(_E2M1_VALUES[:8], dtype=torch.float32,
+                         device=t.device)
+    mag = t.abs().float()
+    idx = (mag.unsqueeze(-1) == table).to(torch.uint8).argmax(-1)
+    ok = table[idx] == mag
+    if not bool(ok.all()):
+        raise ValueError("tensor contains values not representable in e2m1")
+    nib = (idx + 8 * (t < 0).to(idx.dtype)).to(torch.uint8)
+    lo = nib[..., 0::2]
+    hi = nib[..., 1::2]
+    return (lo | (hi << 4)).contiguous()
+
+
+def e2m1_decode(p: torch.Tensor) -> torch.Tensor:
+    """Inverse of e2m1_pack: uint8 nibble-packed -> float32 (last dim
+    doubles)."""
+    table = torch.tensor(_E2M1_VALUES, dtype=torch.float32, device=p.device)
+    lo = table[(p & 0xF).long()]
+    hi = table[(p >> 4).long()]
+    out = torch.stack([lo, hi], dim=-1)
+    return out.view(*p.shape[:-1], p.shape[-1] * 2)
+
+
+def gemm_mxfp4(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
+               a_scale: torch.Tensor, b_scale: torch.Tensor,
+               stream=None, xcd_swizzle: bool = False) -> None:
+    """K7-mx4: block-scaled OCP MX-fp4 GEMM — the 4x-bf16 MFMA rate class.
+
+    C[M,N] fp32 = (A * 2^(As-127)) @ (B * 2^(Bs-127))^T where A/B are
+    nibble-PACKED e2m1 uint8 tensors [M,K//2]/[N,K//2] (low nibble = even
+    k; e2m1_pack produces this layout) and a_scale/b_scale are e8m0
+    exponents [M,K//32]/[N,K//32] as for gemm_mxfp8. The fp4 mode of
+    mfma_scale_f32_16x16x128_f8f6f4; operand/scale layout measured on
+    hardware (diagonal — scripts/probes/fp4_probe*). K = 2*a.shape[1];
+    requires M,N,K % 128 == 0.
+    """
+    if c.dtype != torch.float32 or a.dtype != torch.uint8 \
+            or b.dtype != torch.uint8:
+        raise TypeError("c must be fp32; a, b must be nibble-packed uint8")
+    if a_scale.dtype != torch.uint8 or b_scale.dtype != torch.uint8:
+        raise TypeError("scales must be uint8 (e8m0 exponents)")
+    for t, name in ((c, "c"), (a, "a"), (b, "b"),
+                    (a_scale, "a_scale"), (b_scale, "b_scale")):
+        if not t.is_cuda or not t.is_contiguous() or t.dim() != 2:
+            raise TypeError(f"{name} must be a contiguous 2-D CUDA tensor")
+    m, kb = a.shape
+    n, kb2 = b.shape
+    k = kb * 2
+    if kb2 != kb or c.shape != (m, n):
+        raise ValueError(f"shape mismatch: A{tuple(a.shape)} B{tuple(b.shape)}"
+                         f" C{tuple(c.shape)}")
+    if a_scale.shape != (m, k // 32) or b_scale.shape != (n, k // 32):
+        raise ValueError("scales must be [rows, K//32]")
+    native().gemm_mxfp4_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                           a_scale.data_ptr(), b_scale.data_ptr(),
+                           m, n, k, _stream_handle(stream), int(xcd_swizzle))
+
+
 def gemm_i8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
             stream=None, xcd_swizzle: bool = False) -> None:
     """K7-i8: C[M,N] int32 = A[M,K] @ B[N,K]^T, int8 operands.

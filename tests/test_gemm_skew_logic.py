@@ -150,3 +150,45 @@ def test_group_remap_bijective():
     # the first tiles_m remapped ids all live in columns [0, group)
     first = {remap(w, 32, 32 * 32, 8) % 32 for w in range(8 * 32)}
     assert first == set(range(8))
+
+
+def test_fp4_natural_layout_conflict_free():
+    """The MX-fp4 tile ([128 rows][64 packed bytes], fragment = 16
+    contiguous bytes at (row, 16*g)) needs NO skew: the 64-byte row
+    stride naturally rotates rows across the b128 bank windows. Enumerate
+    the true gfx950 b128 lane groups (as for bf16) and assert all 16
+    lanes of each group hit 16 distinct 16-byte windows."""
+    groups = [
+        [0, 1, 2, 3, 12, 13, 14, 15, 20, 21, 22, 23, 24, 25, 26, 27],
+        [4, 5, 6, 7, 8, 9, 10, 11, 16, 17, 18, 19, 28, 29, 30, 31],
+        [32, 33, 34, 35, 44, 45, 46, 47, 52, 53, 54, 55, 56, 57, 58, 59],
+        [36, 37, 38, 39, 40, 41, 42, 43, 48, 49, 50, 51, 60, 61, 62, 63],
+    ]
+    for mrow_base in (0, 16, 32):  # several fragment row offsets
+        for grp in groups:
+            windows = set()
+            for lane in grp:
+                row = mrow_base + (lane & 15)
+                g = lane >> 4
+                byte = row * 64 + 16 * g
+                windows.add((byte // 16) % 32)
+            assert len(windows) == 16, (mrow_base, grp, sorted(windows))
+
+
+def test_e2m1_pack_decode_roundtrip():
+    import torch
+
+    from hpc_patterns_amd import ops
+
+    vals = torch.tensor([0., 0.5, 1., 1.5, 2., 3., 4., 6.])
+    t = torch.cat([vals, -vals]).repeat(4).view(2, -1)  # [2, 32]
+    p = ops.e2m1_pack(t)
+    assert p.dtype == torch.uint8 and p.shape == (2, 16)
+    d = ops.e2m1_decode(p)
+    assert torch.equal(d.abs(), t.abs())
+    assert bool((d[t != 0] == t[t != 0]).all())
+    import pytest
+    with pytest.raises(ValueError):
+        ops.e2m1_pack(torch.tensor([[5.0, 1.0]]))
+    with pytest.raises(ValueError):
+        ops.e2m1_pack(torch.tensor([[1.0, 2.0, 3.0]]))  # odd last dim

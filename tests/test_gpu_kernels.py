@@ -686,3 +686,102 @@ def test_gemm_mxfp8_grouped_order_exact(monkeypatch):
         ops.gemm_mxfp8(c, a, b, sa, sb)
         torch.cuda.synchronize()
         assert torch.equal(c, ref), grp
+
+
+# ---------------------------------------------------------------------------
+# K7-mx4: block-scaled OCP MX-fp4 GEMM (4x rate class)
+# ---------------------------------------------------------------------------
+
+def _mx4_operands(m, n, k, seed):
+    """Random exactly-representable e2m1 payloads + packed forms."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    from hpc_patterns_amd import ops
+
+    vals = torch.tensor([0., 0.5, -0.5, 1., -1., 1.5, -1.5, 2., -2.])
+    fa = vals[torch.randint(0, 9, (m, k), generator=g)]
+    fb = vals[torch.randint(0, 9, (n, k), generator=g)]
+    return fa, fb, ops.e2m1_pack(fa).cuda(), ops.e2m1_pack(fb).cuda()
+
+
+def test_gemm_mxfp4_unit_scales_exact():
+    """Products/sums of e2m1 values at these magnitudes are exact in fp32:
+    the result must EQUAL the fp32 reference bitwise."""
+    from hpc_patterns_amd import ops
+
+    m, n, k = 128, 256, 384
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 113)
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa, sb)
+    ref = torch.matmul(fa, fb.t()).cuda()
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_mxfp4_identity():
+    """A = packed identity with an ASYMMETRIC B: C = B^T exactly."""
+    from hpc_patterns_amd import ops
+
+    m = n = k = 128
+    fa = torch.eye(m, k)
+    g = torch.Generator(device="cpu").manual_seed(127)
+    vals = torch.tensor([0., 0.5, -0.5, 1., -1., 1.5, -1.5, 2., -2.])
+    fb = vals[torch.randint(0, 9, (n, k), generator=g)]
+    pa, pb = ops.e2m1_pack(fa).cuda(), ops.e2m1_pack(fb).cuda()
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa, sa.clone())
+    torch.cuda.synchronize()
+    assert torch.equal(c, fb.t().cuda()), (c - fb.t().cuda()).abs().max()
+
+
+def test_gemm_mxfp4_power_of_two_scales_exact():
+    """Random e8m0 scales near 127: dequant is exact power-of-2 scaling,
+    still bitwise-equal to the fp32 reference."""
+    from hpc_patterns_amd import ops
+
+    m, n, k = 256, 128, 256
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 131)
+    g = torch.Generator(device="cpu").manual_seed(137)
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    da = fa * torch.pow(2.0, sa.float() - 127).repeat_interleave(32, dim=1)
+    db = fb * torch.pow(2.0, sb.float() - 127).repeat_interleave(32, dim=1)
+    ref = torch.matmul(da, db.t()).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa.cuda(), sb.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_mxfp4_large_square_exact():
+    from hpc_patterns_amd import ops
+
+    m = n = k = 1024
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 139)
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa, sb)
+    ref = torch.matmul(fa, fb.t()).cuda()
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
+
+
+def test_gemm_mxfp4_grouped_order_exact(monkeypatch):
+    from hpc_patterns_amd import ops
+
+    m, n, k = 384, 1152, 256  # 3 x 9 tile grid (partial bands)
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 149)
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    ref = torch.matmul(fa, fb.t()).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    for grp in ("4", "16", "32"):
+        monkeypatch.setenv("HPK_GEMM_GROUP", grp)
+        ops.gemm_mxfp4(c, pa, pb, sa, sb)
+        torch.cuda.synchronize()
+        assert torch.equal(c, ref), grp
