@@ -57,6 +57,8 @@ def main():
     c = torch.empty(sz, sz, dtype=torch.float32, device=dev)
     a8, b8 = a.to(torch.float8_e4m3fn), b.to(torch.float8_e4m3fn)
     s1 = torch.full((sz, sz // 32), 127, dtype=torch.uint8, device=dev)
+    p4a = torch.randint(0, 256, (sz, sz // 2), dtype=torch.uint8, device=dev)
+    p4b = torch.randint(0, 256, (sz, sz // 2), dtype=torch.uint8, device=dev)
     ai = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
     bi = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
     ci = torch.empty(sz, sz, dtype=torch.int32, device=dev)
@@ -73,6 +75,8 @@ def main():
         variants[f"fp8 {tag}"] = (g, lambda: ops.gemm_fp8(c, a8, b8))
         variants[f"mx8 {tag}"] = (g, lambda: ops.gemm_mxfp8(c, a8, b8, s1, s1))
         variants[f"i8 {tag}"] = (g, lambda: ops.gemm_i8(ci, ai, bi))
+        variants[f"mx4 {tag}"] = (g, lambda: ops.gemm_mxfp4(
+            c, p4a, p4b, s1, s1))
     try:
         sa = torch.ones(sz, 1, device=dev)
         sb = torch.ones(1, sz, device=dev)

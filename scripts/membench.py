@@ -116,12 +116,16 @@ def main():
         s1 = torch.full((sz, sz // 32), 127, dtype=torch.uint8, device=dev)
         t = time_gpu(lambda: ops.gemm_mxfp8(c, a8, b8, s1, s1))
         print(f"gemm {sz:5d} mx8  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
+        p4 = torch.randint(0, 256, (sz, sz // 2), dtype=torch.uint8,
+                           device=dev)
+        t = time_gpu(lambda: ops.gemm_mxfp4(c, p4, p4, s1, s1))
+        print(f"gemm {sz:5d} mx4  {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TFLOP/s")
         ai = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
         bi = torch.randint(-128, 128, (sz, sz), dtype=torch.int8, device=dev)
         ci = torch.empty(sz, sz, dtype=torch.int32, device=dev)
         t = time_gpu(lambda: ops.gemm_i8(ci, ai, bi))
         print(f"gemm {sz:5d} i8   {t*1e3:9.3f} ms  {fl/t/1e12:9.1f} TOP/s")
-        del a, b, a8, b8, s1, ai, bi, ci, c
+        del a, b, a8, b8, s1, p4, ai, bi, ci, c
 
 
 if __name__ == "__main__":
