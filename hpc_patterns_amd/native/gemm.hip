@@ -1413,6 +1413,248 @@ void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
 }
 
 
+
+// 4-wave variant: same diagonal-layout staging, 64x64 per wave (MREP =
+// NREP = 4) — 16 MFMAs per wave per K-tile instead of 8, halving the
+// relative cost of the per-tile barrier + staging-drain stall that PMC
+// shows dominating the 8-wave kernel (MfmaUtil 20.6%). HPK_MX4_WAVES
+// selects; the launcher defaults to the measured winner.
+__global__ __launch_bounds__(256) void k_gemm_mxfp4_nt_w4(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int MREP = 4, NREP = 4; // 4 waves as 2x2, 64x64 per wave
+  __shared__ unsigned char lds[2 * 128 * MX4B + 2 * 512];
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 128;
+  const long bcol = (long)(wg % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const int ks = K / 32;
+  const long Kb = (long)K / 2;
+
+  unsigned char* sA = lds + 2 * 128 * MX4B;
+  unsigned char* sB = sA + 512;
+  f32x4 acc[MREP][NREP] = {};
+
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    __syncthreads();
+    // data: [128][64] bytes per operand = 256 threads x 16 B x 2 issues
+    for (int issue = 0; issue < 2; ++issue) {
+      long o_base = (long)issue * 4096 + (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 6);
+      int kk = (int)(o & 63);
+      const unsigned char* ga = A + (brow + row) * Kb + k0 / 2 + kk;
+      const unsigned char* gb = B + (bcol + row) * Kb + k0 / 2 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(lds + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(lds + 128 * MX4B + o_base),
+          16, 0, 0);
+    }
+    // scales: 512 entries per operand, 2 per thread
+    {
+      int e0 = tid * 2;
+      int row = e0 >> 2, kb = e0 & 3;
+      sA[e0] = As[(brow + row) * (long)ks + k0 / 32 + kb];
+      sB[e0] = Bs[(bcol + row) * (long)ks + k0 / 32 + kb];
+      int e1 = e0 + 1;
+      int row1 = e1 >> 2, kb1 = e1 & 3;
+      sA[e1] = As[(brow + row1) * (long)ks + k0 / 32 + kb1];
+      sB[e1] = Bs[(bcol + row1) * (long)ks + k0 / 32 + kb1];
+    }
+    __syncthreads();
+
+    const int g = lane >> 4;
+    typedef __attribute__((ext_vector_type(4))) int i32x4;
+    auto frag16 = [&](const unsigned char* base, long byteoff) {
+      i32x4 lo = *(const i32x4*)__builtin_assume_aligned(base + byteoff, 16);
+      i32x8 f = {};
+      for (int j = 0; j < 4; ++j) f[j] = lo[j];
+      return f;
+    };
+    i32x8 afrag[MREP];
+    int asc[MREP];
+    for (int m = 0; m < MREP; ++m) {
+      int row = wr * 64 + m * 16 + (lane & 15);
+      afrag[m] = frag16(lds, (long)row * MX4B + 16 * g);
+      asc[m] = sA[row * 4 + g];
+    }
+    for (int n = 0; n < NREP; ++n) {
+      int col = wc * 64 + n * 16 + (lane & 15);
+      i32x8 bfrag = frag16(lds + 128 * MX4B, (long)col * MX4B + 16 * g);
+      int bsc = sB[col * 4 + g];
+      for (int m = 0; m < MREP; ++m)
+        acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            afrag[m], bfrag, acc[m][n], 4, 4, 0, asc[m], 0, bsc);
+    }
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * 64 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 64 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
+
+// Double-buffered variant: tile t+1's staging DMAs issue before tile t's
+// compute and a counted s_waitcnt retires exactly tile t's (the bf16 db
+// discipline, findings #18). ALL staging is global_load_lds — including
+// the scales, gathered as 4-byte dwords by waves 0-3 (K % 128 makes the
+// scale rows dword-aligned) — so the per-wave vmcnt count is exact
+// (3 issues for waves 0-3, 2 for waves 4-7; mixing ordinary loads into
+// the queue would break the count — the guide's load-kind trap). PMC
+// motivation: the plain kernel's MfmaUtil is 20.6% — each tile eats a
+// full memory latency inside the barrier window; with 34 KiB LDS and 85
+// VGPRs two 8-wave blocks stay resident on top of the intra-block
+// prefetch. HPK_MX4_WAVES=db selects... (launcher: HPK_GEMM_VARIANT
+// conventions kept: default is the measured winner).
+__global__ __launch_bounds__(512) void k_gemm_mxfp4_nt_db(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int MREP = 4, NREP = 2;
+  constexpr int TILE = 128 * MX4B; // 8 KiB per operand per buffer
+  // ONE shared array (a second __shared__ object would force vmcnt(0)
+  // before every ds_read): [buf][A|B data] x2, then [buf][sA|sB] x2
+  __shared__ unsigned char lds[2 * 2 * TILE + 2 * 1024];
+  unsigned char* const sbase = lds + 2 * 2 * TILE;
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 128;
+  const long bcol = (long)(wg % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+  const int ks = K / 32;
+  const long Kb = (long)K / 2;
+
+  f32x4 acc[MREP][NREP] = {};
+
+  auto stage = [&](int buf, int k0) {
+    unsigned char* dst = lds + (long)buf * 2 * TILE;
+    long o_base = (long)wid * 1024;
+    long o = o_base + (long)lane * 16;
+    int row = (int)(o >> 6);
+    int kk = (int)(o & 63);
+    const unsigned char* ga = A + (brow + row) * Kb + k0 / 2 + kk;
+    const unsigned char* gb = B + (bcol + row) * Kb + k0 / 2 + kk;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)ga,
+        (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gb,
+        (__attribute__((address_space(3))) void*)(dst + TILE + o_base), 16, 0,
+        0);
+    if (wid < 4) {
+      // scale dwords: wave w, lane l gathers rows (w&1)*64+l of As (w<2)
+      // or Bs — dest layout stays s[row*4 + kb]
+      const unsigned char* S = (wid < 2) ? As : Bs;
+      long rbase = (wid < 2) ? brow : bcol;
+      int srow = (wid & 1) * 64 + lane;
+      const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
+      unsigned char* sdst =
+          sbase + (long)buf * 1024 + (wid >= 2 ? 512 : 0) + (wid & 1) * 256;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gs,
+          (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+    }
+  };
+
+  // VALU-diet (PMC showed 4 VALU per MFMA on the lambda version):
+  // per-lane LDS byte offsets are K-invariant — compute once; operand
+  // i32x8s persist across the loop so their top halves (unused in fp4
+  // mode) stay zero with no per-tile re-init/copies.
+  typedef __attribute__((ext_vector_type(4))) int i32x4;
+  const int g = lane >> 4;
+  int a_off[MREP], asc_off[MREP], b_off[NREP], bsc_off[NREP];
+  for (int m = 0; m < MREP; ++m) {
+    int row = wr * 64 + m * 16 + (lane & 15);
+    a_off[m] = row * MX4B + 16 * g;
+    asc_off[m] = row * 4 + g;
+  }
+  for (int n = 0; n < NREP; ++n) {
+    int col = wc * 32 + n * 16 + (lane & 15);
+    b_off[n] = col * MX4B + 16 * g;
+    bsc_off[n] = col * 4 + g;
+  }
+  i32x8 afrag[MREP] = {};
+  i32x8 bfrag[NREP] = {};
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    const int cur = (k0 >> 7) & 1;
+    const bool more = (k0 + 128) < K;
+    if (more) stage(cur ^ 1, k0 + 128);
+    if (more) {
+      if (wid < 4)
+        asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const unsigned char* la = lds + (long)cur * 2 * TILE;
+    const unsigned char* lb = la + TILE;
+    const unsigned char* sA = sbase + (long)cur * 1024;
+    const unsigned char* sB = sA + 512;
+    int asc[MREP];
+    for (int m = 0; m < MREP; ++m) {
+      *(i32x4*)&afrag[m] = *(const i32x4*)__builtin_assume_aligned(
+          la + a_off[m], 16);
+      asc[m] = sA[asc_off[m]];
+    }
+    for (int n = 0; n < NREP; ++n) {
+      *(i32x4*)&bfrag[n] = *(const i32x4*)__builtin_assume_aligned(
+          lb + b_off[n], 16);
+      int bsc = sB[bsc_off[n]];
+      for (int m = 0; m < MREP; ++m)
+        acc[m][n] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            afrag[m], bfrag[n], acc[m][n], 4, 4, 0, asc[m], 0, bsc);
+    }
+    // all waves done reading buf[cur] before the next prefetch overwrites
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int m = 0; m < MREP; ++m)
+    for (int n = 0; n < NREP; ++n) {
+      long row0 = brow + wr * 64 + m * 16 + 4 * (lane >> 4);
+      long col = bcol + wc * 32 + n * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r)
+        C[(row0 + r) * (long)N + col] = acc[m][n][r];
+    }
+}
+
 void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle) {
@@ -1421,10 +1663,30 @@ void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
   const int grp = gemm_group((int)(N / 128));
   int tiles_n = (int)(N / 128);
   int nwg = (int)(M / 128) * tiles_n;
-  hipLaunchKernelGGL(k_gemm_mxfp4_nt, dim3(nwg), dim3(512), 0, stream, C,
-                     (const unsigned char*)A, (const unsigned char*)B,
-                     (const unsigned char*)As, (const unsigned char*)Bs,
-                     (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle, grp);
+  // variants: db (double-buffered, default pending measurement), 8
+  // (plain 8-wave), 4 (plain 4-wave — measured: ties at 8192^3, loses
+  // 12% at 16384^3)
+  const char* v = std::getenv("HPK_MX4_WAVES");
+  std::string waves = v ? v : "db";
+  if (waves == "db") {
+    hipLaunchKernelGGL(k_gemm_mxfp4_nt_db, dim3(nwg), dim3(512), 0, stream,
+                       C, (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle,
+                       grp);
+  } else if (waves == "4") {
+    hipLaunchKernelGGL(k_gemm_mxfp4_nt_w4, dim3(nwg), dim3(256), 0, stream,
+                       C, (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle,
+                       grp);
+  } else {
+    hipLaunchKernelGGL(k_gemm_mxfp4_nt, dim3(nwg), dim3(512), 0, stream, C,
+                       (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tiles_n, nwg, xcd_swizzle,
+                       grp);
+  }
   check_hip(hipGetLastError(), "launch_gemm_mxfp4_nt");
 }
 
