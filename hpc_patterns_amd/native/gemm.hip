@@ -1655,6 +1655,154 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp4_nt_db(
     }
 }
 
+
+// 32x32x64 variant at a 256^2 tile: the bigger scaled MFMA
+// (mfma_scale_f32_32x32x64_f8f6f4, 9099 TF ubench) plus the bigger tile
+// attack the two measured walls in order: an un-skewed 64-byte-stride
+// tile read 8-way bank conflicts (SQ_LDS_BANK_CONFLICT = 4x MFMA count,
+// fixed by the (row>>3)&3 chunk rotation below), and the 128^2 tile's
+// arithmetic intensity (254 FLOP/staged-byte = a ~2.0 PF HBM ceiling at
+// 8 TB/s — the measured 2.0-2.1 PF plateau of ALL 128^2 fp4 variants).
+// 256^2 doubles intensity (508 FLOP/B -> ~4 PF ceiling). fp4_probe3
+// measured the same DIAGONAL operand/scale layout at this shape: lane
+// (row=lane&31, g=lane>>5) supplies OCP block k in [32g,32g+32),
+// own-lane scale byte. 8 waves as 4x2 (64x128 per wave, 16 MFMAs per
+// K-128 tile), f32x16 accumulators (128 AGPRs), double-buffered
+// all-glds staging (scale rows as size-4 glds: sub-dword glds lands in
+// 4-byte-per-lane LDS slots — measured, and a K-128 tile's 4 scale
+// bytes fill the slot exactly).
+__global__ __launch_bounds__(512) void k_gemm_mxfp4_nt_32(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int TILE = 256 * 64; // 16 KiB per operand per buffer (K-128)
+  __shared__ unsigned char lds[2 * 2 * TILE + 2 * 2048];
+  unsigned char* const sbase = lds + 2 * 2 * TILE;
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1; // 4 wave-rows of 64
+  const int wc = wid & 1;  // 2 wave-cols of 128
+  const int ks = K / 32;
+  const long Kb = (long)K / 2;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  typedef __attribute__((ext_vector_type(4))) int i32x4;
+  f32x16 acc[2][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+    unsigned char* dst = lds + (long)buf * 2 * TILE;
+    // anti-bank-conflict chunk rotation: LDS chunk position p of row r
+    // holds global chunk (p - (r>>3)) & 3 (true-lane-group enumeration
+    // in tests/test_gemm_skew_logic.py); glds keeps LDS lane-linear so
+    // the rotation is applied to the SOURCE chunk address
+    for (int issue = 0; issue < 2; ++issue) {
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 6);
+      int p = (int)((o & 63) >> 4);
+      int kk = ((p - (row >> 3)) & 3) * 16;
+      const unsigned char* ga = A + (brow + row) * Kb + k0 / 2 + kk;
+      const unsigned char* gb = B + (bcol + row) * Kb + k0 / 2 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(dst + TILE + o_base), 16,
+          0, 0);
+    }
+    // scales: one dword per row (K-128 tile = 4 e8m0 bytes); waves 0-3
+    // gather As rows wid*64+lane, waves 4-7 Bs
+    const unsigned char* S = (wid < 4) ? As : Bs;
+    long rbase = (wid < 4) ? brow : bcol;
+    int srow = (wid & 3) * 64 + lane;
+    const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
+    unsigned char* sdst =
+        sbase + (long)buf * 2048 + (wid >= 4 ? 1024 : 0) + (wid & 3) * 256;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gs,
+        (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+  };
+
+  // K-invariant per-lane offsets
+  const int g = lane >> 5;
+  const int r31 = lane & 31;
+  int a_off[2], asc_off[2], b_off[4], bsc_off[4];
+  for (int mf = 0; mf < 2; ++mf) {
+    int row = wr * 64 + mf * 32 + r31;
+    a_off[mf] = row * 64;
+    asc_off[mf] = row * 4 + g;
+  }
+  for (int nf = 0; nf < 4; ++nf) {
+    int col = wc * 128 + nf * 32 + r31;
+    b_off[nf] = col * 64;
+    bsc_off[nf] = col * 4 + g;
+  }
+  i32x8 afrag[2] = {};
+  i32x8 bfrag[4] = {};
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    const int cur = (k0 >> 7) & 1;
+    const bool more = (k0 + 128) < K;
+    if (more) stage(cur ^ 1, k0 + 128);
+    if (more)
+      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const unsigned char* la = lds + (long)cur * 2 * TILE;
+    const unsigned char* lb = la + TILE;
+    const unsigned char* sA = sbase + (long)cur * 2048;
+    const unsigned char* sB = sA + 1024;
+    for (int kk = 0; kk < 2; ++kk) {
+      int asc[2];
+      for (int mf = 0; mf < 2; ++mf) {
+        int row = wr * 64 + mf * 32 + r31;
+        int ch = (g + 2 * kk + (row >> 3)) & 3;
+        *(i32x4*)&afrag[mf] = *(const i32x4*)__builtin_assume_aligned(
+            la + a_off[mf] + 16 * ch, 16);
+        asc[mf] = sA[asc_off[mf] + 2 * kk];
+      }
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = wc * 128 + nf * 32 + r31;
+        int ch = (g + 2 * kk + (col >> 3)) & 3;
+        *(i32x4*)&bfrag[nf] = *(const i32x4*)__builtin_assume_aligned(
+            lb + b_off[nf] + 16 * ch, 16);
+        int bsc = sB[bsc_off[nf] + 2 * kk];
+        for (int mf = 0; mf < 2; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              afrag[mf], bfrag[nf], acc[mf][nf], 4, 4, 0, asc[mf], 0, bsc);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int mf = 0; mf < 2; ++mf)
+    for (int nf = 0; nf < 4; ++nf) {
+      long col = bcol + wc * 128 + nf * 32 + r31;
+      for (int r = 0; r < 16; ++r) {
+        long row = brow + wr * 64 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * g;
+        C[row * (long)N + col] = acc[mf][nf][r];
+      }
+    }
+}
+
 void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle) {
@@ -1663,12 +1811,26 @@ void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
   const int grp = gemm_group((int)(N / 128));
   int tiles_n = (int)(N / 128);
   int nwg = (int)(M / 128) * tiles_n;
-  // variants: db (double-buffered, default pending measurement), 8
-  // (plain 8-wave), 4 (plain 4-wave — measured: ties at 8192^3, loses
-  // 12% at 16384^3)
+  // variants: 32 (the 256^2-tile 32x32x64 kernel — DEFAULT for
+  // 256-divisible shapes: 2903/3044 TF at 8192^3/16384^3 vs db 2003),
+  // db (double-buffered 16x16x128 at 128^2, the fallback), 8 (plain
+  // 8-wave), 4 (plain 4-wave — measured negative)
   const char* v = std::getenv("HPK_MX4_WAVES");
-  std::string waves = v ? v : "db";
-  if (waves == "db") {
+  std::string waves = v ? v : "32";
+  if (waves == "32" && M % 256 == 0 && N % 256 == 0) {
+    int tn32 = (int)(N / 256);
+    int n32 = (int)(M / 256) * tn32;
+    // measured (16384^3 interleaved sweep): row-major beats 16/32-wide
+    // bands here — one resident block/CU makes the concurrent footprint
+    // 4 full tile rows, which the MALL already covers
+    const char* genv = std::getenv("HPK_GEMM_GROUP");
+    const int grp32 = genv ? std::atoi(genv) : 1;
+    hipLaunchKernelGGL(k_gemm_mxfp4_nt_32, dim3(n32), dim3(512), 0, stream,
+                       C, (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tn32, n32, xcd_swizzle,
+                       grp32);
+  } else if (waves == "db" || waves == "32") {
     hipLaunchKernelGGL(k_gemm_mxfp4_nt_db, dim3(nwg), dim3(512), 0, stream,
                        C, (const unsigned char*)A, (const unsigned char*)B,
                        (const unsigned char*)As, (const unsigned char*)Bs,

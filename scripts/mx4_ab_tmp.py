@@ -21,11 +21,11 @@ for sz in (8192, 16384):
     c = torch.empty(sz, sz, dtype=torch.float32, device=dev)
     best = {}
     for rnd in range(3):
-        for w in ("8", "4", "db"):
+        for w in ("8", "4", "db", "32"):
             os.environ["HPK_MX4_WAVES"] = w
             best[w] = max(best.get(w, 0),
                           fl/t(lambda: ops.gemm_mxfp4(c, p4a, p4b, s1, s1))/1e12)
-    print(f"{sz}: ctrl={ctrl:.0f}  w8={best[chr(56)]:.0f}  w4={best[chr(52)]:.0f}  db={best[chr(100)+chr(98)]:.0f} TF",
+    print(f"{sz}: ctrl={ctrl:.0f}  w8={best[chr(56)]:.0f}  w4={best[chr(52)]:.0f}  db={best[chr(100)+chr(98)]:.0f}  m32={best[chr(51)+chr(50)]:.0f} TF",
           flush=True)
     del a, b, p4a, p4b, s1, c
     torch.cuda.empty_cache()

@@ -192,3 +192,34 @@ def test_e2m1_pack_decode_roundtrip():
         ops.e2m1_pack(torch.tensor([[5.0, 1.0]]))
     with pytest.raises(ValueError):
         ops.e2m1_pack(torch.tensor([[1.0, 2.0, 3.0]]))  # odd last dim
+
+
+def test_mx4_32x32_chunk_rotation_conflict_free():
+    """The 32x32x64 kernel's 64-byte-stride rows 8-way-conflict without a
+    skew (two rows per (4*row mod 32) window band in every true lane
+    group); the (row>>3)&3 chunk rotation separates them. Enumerate the
+    b128 lane groups over all fragment windows and k-steps."""
+    groups = [
+        [0, 1, 2, 3, 12, 13, 14, 15, 20, 21, 22, 23, 24, 25, 26, 27],
+        [4, 5, 6, 7, 8, 9, 10, 11, 16, 17, 18, 19, 28, 29, 30, 31],
+        [32, 33, 34, 35, 44, 45, 46, 47, 52, 53, 54, 55, 56, 57, 58, 59],
+        [36, 37, 38, 39, 40, 41, 42, 43, 48, 49, 50, 51, 60, 61, 62, 63],
+    ]
+    for base in (0, 32, 64, 96, 128, 224):  # fragment 32-row windows
+        for kk in (0, 1):
+            for grp in groups:
+                windows = set()
+                for lane in grp:
+                    row = base + (lane & 31)
+                    g = lane >> 5
+                    ch = (g + 2 * kk + (row >> 3)) & 3
+                    byte = row * 64 + 16 * ch
+                    windows.add((byte // 16) % 32)
+                assert len(windows) == 16, (base, kk, grp, sorted(windows))
+    # and WITHOUT the rotation it is genuinely 8-way broken
+    bad = set()
+    for lane in groups[0]:
+        row = lane & 31
+        byte = row * 64 + 16 * (lane >> 5)
+        bad.add((byte // 16) % 32)
+    assert len(bad) < 16

@@ -785,3 +785,25 @@ def test_gemm_mxfp4_grouped_order_exact(monkeypatch):
         ops.gemm_mxfp4(c, pa, pb, sa, sb)
         torch.cuda.synchronize()
         assert torch.equal(c, ref), grp
+
+
+def test_gemm_mxfp4_256tile_power_of_two_scales_exact():
+    """256-divisible shapes take the 256^2-tile 32x32x64 kernel (the
+    default fast path): full exactness incl. its size-4 glds scale
+    staging with non-uniform scales."""
+    from hpc_patterns_amd import ops
+
+    m, n, k = 256, 512, 384
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 151)
+    g = torch.Generator(device="cpu").manual_seed(157)
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    da = fa * torch.pow(2.0, sa.float() - 127).repeat_interleave(32, dim=1)
+    db = fb * torch.pow(2.0, sb.float() - 127).repeat_interleave(32, dim=1)
+    ref = torch.matmul(da, db.t()).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa.cuda(), sb.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
