@@ -235,10 +235,13 @@ def gemm_mxfp4(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
     C[M,N] fp32 = (A * 2^(As-127)) @ (B * 2^(Bs-127))^T where A/B are
     nibble-PACKED e2m1 uint8 tensors [M,K//2]/[N,K//2] (low nibble = even
     k; e2m1_pack produces this layout) and a_scale/b_scale are e8m0
-    exponents [M,K//32]/[N,K//32] as for gemm_mxfp8. The fp4 mode of
-    mfma_scale_f32_16x16x128_f8f6f4; operand/scale layout measured on
-    hardware (diagonal — scripts/probes/fp4_probe*). K = 2*a.shape[1];
-    requires M,N,K % 128 == 0.
+    exponents [M,K//32]/[N,K//32] as for gemm_mxfp8. The fp4 mode of the
+    scaled MFMAs; operand/scale layout measured on hardware (diagonal —
+    scripts/probes/fp4_probe*). Default kernel: the 256^2-tile
+    double-buffered 32x32x64 design, 2903-3197 TF at 8192^3-16384^3
+    (3.2 PFLOP/s; HPK_MX4_WAVES=db|8|4 selects the 128^2 variants, used
+    automatically when M or N is not a multiple of 256). K =
+    2*a.shape[1]; requires M,N,K % 128 == 0.
     """
     if c.dtype != torch.float32 or a.dtype != torch.uint8 \
             or b.dtype != torch.uint8:
