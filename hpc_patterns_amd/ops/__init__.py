@@ -291,10 +291,11 @@ def gemm_pad_shapes(kind: str, m: int, n: int, k: int):
     """Padded (M, N, K) that puts a (m, n, k) problem on the fast path.
 
     kind is "bf16" / "fp8" (8-phase needs M,N % 256, K % 128), "i8"
-    (M,N % 256, K % 256) or "mxfp8" (plain kernel only: M,N,K % 128).
+    (M,N % 256, K % 256), "mxfp8" (plain kernel only: M,N,K % 128) or
+    "mxfp4" (the 256^2 32x32x64 kernel: M,N % 256, K % 128).
     Pure shape math — unit-tested on CPU (tests/test_gemm_skew_logic.py).
     """
-    if kind not in ("bf16", "fp8", "i8", "mxfp8"):
+    if kind not in ("bf16", "fp8", "i8", "mxfp8", "mxfp4"):
         raise ValueError(f"unknown gemm kind {kind!r}")
     def up(x, q):
         return -(-x // q) * q
@@ -319,7 +320,8 @@ def matmul_nt(a: torch.Tensor, b: torch.Tensor,
     """Arbitrary-shape front door to the K7 GEMM family: returns A @ B^T.
 
     Dispatches on dtype — bf16 -> gemm_bf16, float8_e4m3fn -> gemm_fp8
-    (or gemm_mxfp8 when e8m0 scales are given), int8 -> gemm_i8 — after
+    (or gemm_mxfp8 when e8m0 scales are given), int8 -> gemm_i8, uint8 +
+    scales -> gemm_mxfp4 (nibble-packed e2m1; shapes in elements) — after
     zero-padding the operands up to the fast-path tile multiples
     (gemm_pad_shapes). Zero rows/columns contribute nothing, so the
     result is bit-identical to the unpadded kernel output; the padded
@@ -341,9 +343,32 @@ def matmul_nt(a: torch.Tensor, b: torch.Tensor,
             kind, fn, out_dtype = "mxfp8", None, torch.float32
         else:
             kind, fn, out_dtype = "fp8", gemm_fp8, torch.float32
+    elif a.dtype == torch.uint8:
+        # nibble-packed e2m1 (e2m1_pack layout): shapes are in ELEMENTS,
+        # K = 2 * packed columns; zero nibbles pad exactly (+0.0)
+        if a_scale is None:
+            raise TypeError("packed-fp4 operands need e8m0 scales")
+        kind, fn, out_dtype = "mxfp4", None, torch.float32
+        k = 2 * k
     else:
         raise TypeError(f"unsupported operand dtype {a.dtype}")
     mp, np_, kp = gemm_pad_shapes(kind, m, n, k)
+    if kind == "mxfp4":
+        if b_scale is None or a_scale.shape != (m, k // 32) \
+                or b_scale.shape != (n, k // 32):
+            raise ValueError("mx4 path needs a_scale [M,K//32] and "
+                             "b_scale [N,K//32] (K = 2*packed cols, "
+                             "K % 32 == 0)")
+        ap = _pad2d(a.contiguous(), mp, kp // 2)
+        bp = _pad2d(b.contiguous(), np_, kp // 2)
+        c = torch.empty(mp, np_, dtype=out_dtype, device=a.device)
+        asp = _pad2d(a_scale.contiguous(), mp, kp // 32, fill=127)
+        bsp = _pad2d(b_scale.contiguous(), np_, kp // 32, fill=127)
+        gemm_mxfp4(c, ap, bp, asp, bsp, stream=stream,
+                   xcd_swizzle=xcd_swizzle)
+        if (mp, np_) == (m, n):
+            return c
+        return c[:m, :n].contiguous()
     ap = _pad2d(a.contiguous(), mp, kp)
     bp = _pad2d(b.contiguous(), np_, kp)
     c = torch.empty(mp, np_, dtype=out_dtype, device=a.device)

@@ -223,3 +223,12 @@ def test_mx4_32x32_chunk_rotation_conflict_free():
         byte = row * 64 + 16 * (lane >> 5)
         bad.add((byte // 16) % 32)
     assert len(bad) < 16
+
+
+def test_gemm_pad_shapes_mxfp4():
+    from hpc_patterns_amd.ops import gemm_pad_shapes
+
+    assert gemm_pad_shapes("mxfp4", 256, 512, 128) == (256, 512, 128)
+    assert gemm_pad_shapes("mxfp4", 200, 136, 192) == (256, 256, 256)
+    mp, np_, kp = gemm_pad_shapes("mxfp4", 1, 1, 32)
+    assert mp % 256 == 0 and np_ % 256 == 0 and kp % 128 == 0

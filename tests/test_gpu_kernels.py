@@ -807,3 +807,18 @@ def test_gemm_mxfp4_256tile_power_of_two_scales_exact():
     ops.gemm_mxfp4(c, pa, pb, sa.cuda(), sb.cuda())
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_matmul_nt_mxfp4_odd_shapes_exact():
+    """Packed-fp4 through the matmul_nt padding path: zero nibbles pad
+    exactly, scales pad with 127."""
+    from hpc_patterns_amd import ops
+
+    m, n, k = 200, 136, 192  # odd vs the 256/128 fast-path multiples
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 167)
+    sa = torch.full((m, k // 32), 127, dtype=torch.uint8, device="cuda")
+    sb = torch.full((n, k // 32), 127, dtype=torch.uint8, device="cuda")
+    c = ops.matmul_nt(pa, pb, a_scale=sa, b_scale=sb)
+    ref = torch.matmul(fa, fb.t()).cuda()
+    torch.cuda.synchronize()
+    assert c.shape == (m, n) and torch.equal(c, ref)
