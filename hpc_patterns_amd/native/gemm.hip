@@ -1399,11 +1399,173 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp4_nt(
     }
 }
 
+
+// 256^2-tile 32x32x64 MX-fp8 kernel — the mx4 _32 design (see
+// k_gemm_mxfp4_nt_32) ported to fp8: same arithmetic-intensity argument
+// (the 128^2 mx8 kernel's 127 FLOP/staged-byte = a ~1 PF naive HBM
+// ceiling, measured 1.4-1.5 PF with LLC help; 256^2 doubles it), same
+// double-buffered all-glds staging. fp8 differences, both measured on
+// hardware (scripts/probes/fp8_probe*_32x32): the operand/scale layout
+// at this shape is HALF-INTERLEAVED, not diagonal — scale lane (row,gs)
+// covers bytes [16gs,+16) of BOTH g-lanes, so half h of lane (row,g)
+// feeds from k [32h + 16g, +16) and the lane passes the scale byte for
+// block g — and the 128-byte rows need a (row>>2)&7 chunk rotation
+// (4 rows share each (8row mod 32) window band; their row>>2 values are
+// distinct mod 8 — enumerated in tests/test_gemm_skew_logic.py).
+__global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int TILE = 256 * 128; // 32 KiB per operand per buffer (K-128)
+  __shared__ unsigned char lds[2 * 2 * TILE + 2 * 2048];
+  unsigned char* const sbase = lds + 2 * 2 * TILE;
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const int ks = K / 32;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  typedef __attribute__((ext_vector_type(4))) int i32x4;
+  f32x16 acc[2][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+    unsigned char* dst = lds + (long)buf * 2 * TILE;
+    for (int issue = 0; issue < 4; ++issue) {
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 7);
+      int p = (int)((o & 127) >> 4);
+      int kk = ((p - (row >> 2)) & 7) * 16;
+      const unsigned char* ga = A + (brow + row) * (long)K + k0 + kk;
+      const unsigned char* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(dst + TILE + o_base), 16,
+          0, 0);
+    }
+    const unsigned char* S = (wid < 4) ? As : Bs;
+    long rbase = (wid < 4) ? brow : bcol;
+    int srow = (wid & 3) * 64 + lane;
+    const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
+    unsigned char* sdst =
+        sbase + (long)buf * 2048 + (wid >= 4 ? 1024 : 0) + (wid & 3) * 256;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gs,
+        (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+  };
+
+  const int g = lane >> 5;
+  const int r31 = lane & 31;
+  int a_off[2], asc_off[2], b_off[4], bsc_off[4];
+  for (int mf = 0; mf < 2; ++mf) {
+    int row = wr * 64 + mf * 32 + r31;
+    a_off[mf] = row * 128;
+    asc_off[mf] = row * 4 + g;
+  }
+  for (int nf = 0; nf < 4; ++nf) {
+    int col = wc * 128 + nf * 32 + r31;
+    b_off[nf] = col * 128;
+    bsc_off[nf] = col * 4 + g;
+  }
+  i32x8 afrag[2];
+  i32x8 bfrag[4];
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    const int cur = (k0 >> 7) & 1;
+    const bool more = (k0 + 128) < K;
+    if (more) stage(cur ^ 1, k0 + 128);
+    if (more)
+      asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const unsigned char* la = lds + (long)cur * 2 * TILE;
+    const unsigned char* lb = la + TILE;
+    const unsigned char* sA = sbase + (long)cur * 2048;
+    const unsigned char* sB = sA + 1024;
+    for (int kk = 0; kk < 2; ++kk) {
+      int asc[2];
+      for (int mf = 0; mf < 2; ++mf) {
+        int row = wr * 64 + mf * 32 + r31;
+        int rot = row >> 2;
+        int ch0 = (4 * kk + g + rot) & 7;      // half h=0
+        int ch1 = (4 * kk + 2 + g + rot) & 7;  // half h=1
+        *(i32x4*)&afrag[mf] = *(const i32x4*)__builtin_assume_aligned(
+            la + a_off[mf] + 16 * ch0, 16);
+        *((i32x4*)&afrag[mf] + 1) = *(const i32x4*)__builtin_assume_aligned(
+            la + a_off[mf] + 16 * ch1, 16);
+        asc[mf] = sA[asc_off[mf] + 2 * kk];
+      }
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = wc * 128 + nf * 32 + r31;
+        int rot = col >> 2;
+        int ch0 = (4 * kk + g + rot) & 7;
+        int ch1 = (4 * kk + 2 + g + rot) & 7;
+        *(i32x4*)&bfrag[nf] = *(const i32x4*)__builtin_assume_aligned(
+            lb + b_off[nf] + 16 * ch0, 16);
+        *((i32x4*)&bfrag[nf] + 1) = *(const i32x4*)__builtin_assume_aligned(
+            lb + b_off[nf] + 16 * ch1, 16);
+        int bsc = sB[bsc_off[nf] + 2 * kk];
+        for (int mf = 0; mf < 2; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0, asc[mf], 0, bsc);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int mf = 0; mf < 2; ++mf)
+    for (int nf = 0; nf < 4; ++nf) {
+      long col = bcol + wc * 128 + nf * 32 + r31;
+      for (int r = 0; r < 16; ++r) {
+        long row = brow + wr * 64 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * g;
+        C[row * (long)N + col] = acc[mf][nf][r];
+      }
+    }
+}
+
 void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle) {
   if (M % 128 != 0 || N % 128 != 0 || K % 128 != 0)
     throw std::runtime_error("gemm_mxfp8_nt requires M,N,K % 128 == 0");
+  // default: the 256^2 32x32x64 kernel for eligible shapes
+  // (HPK_MX8_VARIANT=plain forces the 128^2 16x16x128 kernel)
+  const char* v8 = std::getenv("HPK_MX8_VARIANT");
+  const bool m32 = (!v8 || std::string(v8) == "32");
+  if (m32 && M % 256 == 0 && N % 256 == 0) {
+    int tn32 = (int)(N / 256);
+    int n32 = (int)(M / 256) * tn32;
+    const char* genv = std::getenv("HPK_GEMM_GROUP");
+    const int grp32 = genv ? std::atoi(genv) : 1; // row-major (as mx4 _32)
+    hipLaunchKernelGGL(k_gemm_mxfp8_nt_32, dim3(n32), dim3(512), 0, stream,
+                       C, (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tn32, n32, xcd_swizzle,
+                       grp32);
+    check_hip(hipGetLastError(), "launch_gemm_mxfp8_nt(32)");
+    return;
+  }
   const int grp = gemm_group((int)(N / 128));
   int tiles_n = (int)(N / 128);
   int nwg = (int)(M / 128) * tiles_n;

@@ -232,3 +232,33 @@ def test_gemm_pad_shapes_mxfp4():
     assert gemm_pad_shapes("mxfp4", 200, 136, 192) == (256, 256, 256)
     mp, np_, kp = gemm_pad_shapes("mxfp4", 1, 1, 32)
     assert mp % 256 == 0 and np_ % 256 == 0 and kp % 128 == 0
+
+
+def test_mx8_32x32_chunk_rotation_conflict_free():
+    """The fp8 32x32x64 kernel's 128-byte rows put FOUR rows in each
+    (8*row mod 32) window band per true lane group; the (row>>2)&7 chunk
+    rotation separates them (their row>>2 values are distinct mod 8)."""
+    groups = [
+        [0, 1, 2, 3, 12, 13, 14, 15, 20, 21, 22, 23, 24, 25, 26, 27],
+        [4, 5, 6, 7, 8, 9, 10, 11, 16, 17, 18, 19, 28, 29, 30, 31],
+        [32, 33, 34, 35, 44, 45, 46, 47, 52, 53, 54, 55, 56, 57, 58, 59],
+        [36, 37, 38, 39, 40, 41, 42, 43, 48, 49, 50, 51, 60, 61, 62, 63],
+    ]
+    for base in (0, 32, 96, 224):
+        for kk in (0, 1):
+            for h in (0, 1):
+                for grp in groups:
+                    windows = set()
+                    for lane in grp:
+                        row = base + (lane & 31)
+                        g = lane >> 5
+                        ch = (4 * kk + 2 * h + g + (row >> 2)) & 7
+                        byte = row * 128 + 16 * ch
+                        windows.add((byte // 16) % 32)
+                    assert len(windows) == 16, (base, kk, h, sorted(windows))
+    # without the rotation: 4-way broken
+    bad = set()
+    for lane in groups[0]:
+        byte = (lane & 31) * 128 + 16 * (lane >> 5)
+        bad.add((byte // 16) % 32)
+    assert len(bad) <= 8

@@ -822,3 +822,27 @@ def test_matmul_nt_mxfp4_odd_shapes_exact():
     ref = torch.matmul(fa, fb.t()).cuda()
     torch.cuda.synchronize()
     assert c.shape == (m, n) and torch.equal(c, ref)
+
+
+def test_gemm_mxfp8_256tile_power_of_two_scales_exact():
+    """256-divisible shapes take the 256^2-tile 32x32x64 mx8 kernel (the
+    r2-follow-up default): full exactness incl. non-uniform scales."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(173)
+    m, n, k = 256, 512, 384
+    fa = torch.randint(-4, 5, (m, k), generator=g).float()
+    fb = torch.randint(-4, 5, (n, k), generator=g).float()
+    a = fa.to(torch.float8_e4m3fn).cuda()
+    b = fb.to(torch.float8_e4m3fn).cuda()
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    da = fa * torch.pow(2.0, sa.float() - 127).repeat_interleave(32, dim=1)
+    db = fb * torch.pow(2.0, sb.float() - 127).repeat_interleave(32, dim=1)
+    ref = torch.matmul(da, db.t()).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp8(c, a, b, sa.cuda(), sb.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
