@@ -1412,6 +1412,11 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp4_nt(
 // block g — and the 128-byte rows need a (row>>2)&7 chunk rotation
 // (4 rows share each (8row mod 32) window band; their row>>2 values are
 // distinct mod 8 — enumerated in tests/test_gemm_skew_logic.py).
+// SCALED=false turns it into the plain-fp8 kernel: the scaled MFMA with
+// a hardcoded 127 (= x1.0) scale IS the non-scaled fp8 GEMM, and no
+// 32x32x64 non-scaled fp8 MFMA exists — scale staging drops out
+// entirely (one fewer glds per wave per tile).
+template <bool SCALED>
 __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
     float* __restrict__ C, const unsigned char* __restrict__ A,
     const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
@@ -1459,15 +1464,17 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
           (__attribute__((address_space(3))) void*)(dst + TILE + o_base), 16,
           0, 0);
     }
-    const unsigned char* S = (wid < 4) ? As : Bs;
-    long rbase = (wid < 4) ? brow : bcol;
-    int srow = (wid & 3) * 64 + lane;
-    const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
-    unsigned char* sdst =
-        sbase + (long)buf * 2048 + (wid >= 4 ? 1024 : 0) + (wid & 3) * 256;
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)gs,
-        (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+    if (SCALED) {
+      const unsigned char* S = (wid < 4) ? As : Bs;
+      long rbase = (wid < 4) ? brow : bcol;
+      int srow = (wid & 3) * 64 + lane;
+      const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
+      unsigned char* sdst =
+          sbase + (long)buf * 2048 + (wid >= 4 ? 1024 : 0) + (wid & 3) * 256;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gs,
+          (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+    }
   };
 
   const int g = lane >> 5;
@@ -1492,7 +1499,7 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
     const bool more = (k0 + 128) < K;
     if (more) stage(cur ^ 1, k0 + 128);
     if (more)
-      asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(SCALED ? 9 : 8) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1513,7 +1520,7 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
             la + a_off[mf] + 16 * ch0, 16);
         *((i32x4*)&afrag[mf] + 1) = *(const i32x4*)__builtin_assume_aligned(
             la + a_off[mf] + 16 * ch1, 16);
-        asc[mf] = sA[asc_off[mf] + 2 * kk];
+        asc[mf] = SCALED ? sA[asc_off[mf] + 2 * kk] : 127;
       }
       for (int nf = 0; nf < 4; ++nf) {
         int col = wc * 128 + nf * 32 + r31;
@@ -1524,7 +1531,7 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp8_nt_32(
             lb + b_off[nf] + 16 * ch0, 16);
         *((i32x4*)&bfrag[nf] + 1) = *(const i32x4*)__builtin_assume_aligned(
             lb + b_off[nf] + 16 * ch1, 16);
-        int bsc = sB[bsc_off[nf] + 2 * kk];
+        int bsc = SCALED ? sB[bsc_off[nf] + 2 * kk] : 127;
         for (int mf = 0; mf < 2; ++mf)
           acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0, asc[mf], 0, bsc);
@@ -1558,7 +1565,8 @@ void launch_gemm_mxfp8_nt(float* C, const void* A, const void* B,
     int n32 = (int)(M / 256) * tn32;
     const char* genv = std::getenv("HPK_GEMM_GROUP");
     const int grp32 = genv ? std::atoi(genv) : 1; // row-major (as mx4 _32)
-    hipLaunchKernelGGL(k_gemm_mxfp8_nt_32, dim3(n32), dim3(512), 0, stream,
+    hipLaunchKernelGGL((k_gemm_mxfp8_nt_32<true>), dim3(n32), dim3(512), 0,
+                       stream,
                        C, (const unsigned char*)A, (const unsigned char*)B,
                        (const unsigned char*)As, (const unsigned char*)Bs,
                        (int)M, (int)N, (int)K, tn32, n32, xcd_swizzle,
@@ -2025,6 +2033,21 @@ void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
   const int grp = gemm_group((int)(N / 256));
   const char* var = std::getenv("HPK_GEMM_VARIANT");
   const bool ph8 = !var || std::string(var) == "8ph";
+  // default: the 256^2 32x32x64 scaled-MFMA kernel with hardcoded x1.0
+  // scales (no non-scaled 32x32x64 fp8 MFMA exists) — measured above the
+  // 16x16x128 8-phase pipeline; HPK_GEMM_VARIANT=8ph|plain|db forces it off
+  if (!var && M % 256 == 0 && N % 256 == 0 && K % 128 == 0) {
+    int tn32 = (int)(N / 256);
+    int n32 = (int)(M / 256) * tn32;
+    const char* genv = std::getenv("HPK_GEMM_GROUP");
+    const int grp32 = genv ? std::atoi(genv) : 1;
+    hipLaunchKernelGGL((k_gemm_mxfp8_nt_32<false>), dim3(n32), dim3(512), 0,
+                       stream, C, (const unsigned char*)A,
+                       (const unsigned char*)B, nullptr, nullptr, (int)M,
+                       (int)N, (int)K, tn32, n32, xcd_swizzle, grp32);
+    check_hip(hipGetLastError(), "launch_gemm_fp8_nt(32)");
+    return;
+  }
   if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0) {
     int tn = (int)(N / 256);
     int n8 = (int)(M / 256) * tn;

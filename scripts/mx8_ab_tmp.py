@@ -24,7 +24,14 @@ for sz in (8192, 16384):
             os.environ["HPK_MX8_VARIANT"] = w
             best[w] = max(best.get(w, 0),
                           fl/t(lambda: ops.gemm_mxfp8(c, a8, b8, s1, s1))/1e12)
-    print(f"{sz}: ctrl={ctrl:.0f}  plain={best['plain']:.0f}  "
-          f"m32={best['32']:.0f} TF", flush=True)
+        os.environ["HPK_GEMM_VARIANT"] = "8ph"
+        best["fp8_8ph"] = max(best.get("fp8_8ph", 0),
+                              fl/t(lambda: ops.gemm_fp8(c, a8, b8))/1e12)
+        os.environ.pop("HPK_GEMM_VARIANT", None)
+        best["fp8_32"] = max(best.get("fp8_32", 0),
+                             fl/t(lambda: ops.gemm_fp8(c, a8, b8))/1e12)
+    print(f"{sz}: ctrl={ctrl:.0f}  mx8plain={best['plain']:.0f}  "
+          f"mx8_32={best['32']:.0f}  fp8_8ph={best['fp8_8ph']:.0f}  "
+          f"fp8_32={best['fp8_32']:.0f} TF", flush=True)
     del a, b, a8, b8, s1, c
     torch.cuda.empty_cache()

@@ -441,9 +441,12 @@ def test_gemm_bf16_8phase_exact_integers():
     assert torch.equal(ci, bi.t().float())
 
 
-def test_gemm_fp8_8phase_exact_integers():
-    """fp8 8-phase default path (M,N%256, K%128): bitwise equality."""
+def test_gemm_fp8_8phase_exact_integers(monkeypatch):
+    """fp8 8-phase pipeline (pinned via HPK_GEMM_VARIANT — 256-divisible
+    shapes now default to the 32x32x64 kernel): bitwise equality."""
     from hpc_patterns_amd import ops
+
+    monkeypatch.setenv("HPK_GEMM_VARIANT", "8ph")
 
     g = torch.Generator(device="cpu").manual_seed(37)
     m, n, k = 256, 512, 640
@@ -844,5 +847,21 @@ def test_gemm_mxfp8_256tile_power_of_two_scales_exact():
     ref = torch.matmul(da, db.t()).cuda()
     c = torch.empty(m, n, dtype=torch.float32, device="cuda")
     ops.gemm_mxfp8(c, a, b, sa.cuda(), sb.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_fp8_32x32_default_exact():
+    """Plain fp8 at 256-divisible shapes defaults to the 256^2 32x32x64
+    scaled-MFMA kernel with hardcoded x1.0 scales: bitwise equality."""
+    from hpc_patterns_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(179)
+    m, n, k = 256, 512, 640
+    a = torch.randint(-3, 4, (m, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    b = torch.randint(-3, 4, (n, k), generator=g).to(torch.float8_e4m3fn).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_fp8(c, a, b)
+    ref = torch.matmul(a.float(), b.float().t())
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
