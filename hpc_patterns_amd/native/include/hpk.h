@@ -72,8 +72,9 @@ void launch_acc_f32_nt(float* dst, const float* src, size_t n,
 void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
                          long N, long K, hipStream_t stream,
                          int xcd_swizzle = 1);
-// fp8 (OCP e4m3) twin — the mfma_f32_16x16x32_fp8_fp8 path: same MFMA
-// rate as bf16 with half the staging bytes.
+// fp8 (OCP e4m3): 256-divisible shapes default to the 256^2 32x32x64
+// scaled-MFMA kernel with hardcoded x1.0 scales (2.2 PF); otherwise the
+// mfma_f32_16x16x32_fp8_fp8 family.
 void launch_gemm_fp8_nt(float* C, const void* A, const void* B, long M,
                         long N, long K, hipStream_t stream,
                         int xcd_swizzle = 0);

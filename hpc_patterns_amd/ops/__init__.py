@@ -140,9 +140,12 @@ def gemm_fp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
              stream=None, xcd_swizzle: bool = False) -> None:
     """K7-fp8: C[M,N] fp32 = A[M,K] @ B[N,K]^T, A/B torch.float8_e4m3fn.
 
-    The mfma_f32_16x16x32_fp8_fp8 twin of gemm_bf16 (same MFMA rate, half
-    the staging bytes; gfx950 fp8 is OCP e4m3 — exactly torch's
-    float8_e4m3fn). Same shape constraints as gemm_bf16.
+    gfx950 fp8 is OCP e4m3 — exactly torch's float8_e4m3fn. Default for
+    256-divisible shapes: the 256^2-tile 32x32x64 scaled-MFMA kernel with
+    hardcoded x1.0 scales (2154-2194 TF — no non-scaled 32x32x64 fp8 MFMA
+    exists); otherwise the mfma_f32_16x16x32_fp8_fp8 family
+    (HPK_GEMM_VARIANT=8ph|plain|db selects). Same shape constraints as
+    gemm_bf16.
     """
     if c.dtype != torch.float32 or a.dtype != torch.float8_e4m3fn \
             or b.dtype != torch.float8_e4m3fn:
@@ -162,7 +165,9 @@ def gemm_fp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
 def gemm_mxfp8(c: torch.Tensor, a: torch.Tensor, b: torch.Tensor,
                a_scale: torch.Tensor, b_scale: torch.Tensor,
                stream=None, xcd_swizzle: bool = False) -> None:
-    """K7-mx: block-scaled MX-fp8 GEMM at 2x the bf16 MFMA rate.
+    """K7-mx: block-scaled MX-fp8 GEMM (the 2x-bf16 rate class;
+    1824-1863 TF via the default 256^2-tile 32x32x64 kernel for
+    256-divisible shapes, HPK_MX8_VARIANT=plain selects the 128^2 one).
 
     C[M,N] fp32 = (A * 2^(As-127)) @ (B * 2^(Bs-127))^T — A/B are
     float8_e4m3fn [M,K]/[N,K]; a_scale/b_scale are uint8 e8m0 exponents
