@@ -66,8 +66,10 @@ void launch_acc_f32_nt(float* dst, const float* src, size_t n,
 
 // K7 (r2, beyond-parity showcase): LDS-tiled bf16 MFMA GEMM —
 // C[M,N] (fp32) = A[M,K] x B[N,K]^T, both operands bf16 K-contiguous.
-// 128x128 tile / 4 waves / v_mfma_f32_16x16x32_bf16, one-buffer
-// global_load_lds staging, optional bijective XCD workgroup swizzle.
+// Default for M,N % 256 / K % 128: the 256^2-tile 8-phase deep pipeline
+// (v_mfma_f32_16x16x32_bf16, counted vmcnt, zero bank conflicts);
+// otherwise the plain/db 128^2 kernels (HPK_GEMM_VARIANT /
+// HPK_GEMM_WAVES select). Optional bijective XCD workgroup swizzle.
 // Requires M,N % 128 == 0 and K % 64 == 0 (throws otherwise).
 void launch_gemm_bf16_nt(float* C, const void* A, const void* B, long M,
                          long N, long K, hipStream_t stream,
