@@ -1261,6 +1261,112 @@ __global__ __launch_bounds__(512) void k_gemm_i8_8ph(
 // from interleaved A/B sweeps at 8192^3 and 16384^3 (clock-drift-controlled;
 // docs/gemm.md): row-major is already LLC-friendly up to ~32 N-tiles, wider
 // grids gain 4-7% from 16/32-wide bands. HPK_GEMM_GROUP overrides.
+
+// 256^2-tile 32x32x32 int8 kernel — the fp8 _32 design (see
+// k_gemm_mxfp8_nt_32) at the non-scaled i8 MFMA: per-lane fragment = 16
+// contiguous K elements at k = 16*(lane>>5) of each K-32 step (one
+// ds_read_b128), exact int32 accumulation, no scale staging (vmcnt(8)
+// prefetch count), same (row>>2)&7 anti-conflict chunk rotation on the
+// 128-byte rows.
+__global__ __launch_bounds__(512) void k_gemm_i8_nt_32(
+    int* __restrict__ C, const signed char* __restrict__ A,
+    const signed char* __restrict__ B, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int TILE = 256 * 128; // 32 KiB per operand per buffer (K-128)
+  __shared__ unsigned char lds[2 * 2 * TILE];
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+
+  typedef __attribute__((ext_vector_type(16))) int i32x16;
+  typedef __attribute__((ext_vector_type(4))) int i32x4;
+  i32x16 acc[2][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+    unsigned char* dst = lds + (long)buf * 2 * TILE;
+    for (int issue = 0; issue < 4; ++issue) {
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 7);
+      int p = (int)((o & 127) >> 4);
+      int kk = ((p - (row >> 2)) & 7) * 16;
+      const signed char* ga = A + (brow + row) * (long)K + k0 + kk;
+      const signed char* gb = B + (bcol + row) * (long)K + k0 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(dst + TILE + o_base), 16,
+          0, 0);
+    }
+  };
+
+  const int g = lane >> 5;
+  const int r31 = lane & 31;
+  int a_off[2], b_off[4];
+  for (int mf = 0; mf < 2; ++mf) a_off[mf] = (wr * 64 + mf * 32 + r31) * 128;
+  for (int nf = 0; nf < 4; ++nf) b_off[nf] = (wc * 128 + nf * 32 + r31) * 128;
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    const int cur = (k0 >> 7) & 1;
+    const bool more = (k0 + 128) < K;
+    if (more) stage(cur ^ 1, k0 + 128);
+    if (more)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const unsigned char* la = lds + (long)cur * 2 * TILE;
+    const unsigned char* lb = la + TILE;
+    for (int kk = 0; kk < 4; ++kk) {
+      i32x4 afrag[2];
+      for (int mf = 0; mf < 2; ++mf) {
+        int row = wr * 64 + mf * 32 + r31;
+        int ch = (2 * kk + g + (row >> 2)) & 7;
+        afrag[mf] = *(const i32x4*)__builtin_assume_aligned(
+            la + a_off[mf] + 16 * ch, 16);
+      }
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = wc * 128 + nf * 32 + r31;
+        int ch = (2 * kk + g + (col >> 2)) & 7;
+        i32x4 bfrag = *(const i32x4*)__builtin_assume_aligned(
+            lb + b_off[nf] + 16 * ch, 16);
+        for (int mf = 0; mf < 2; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_i32_32x32x32_i8(
+              afrag[mf], bfrag, acc[mf][nf], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int mf = 0; mf < 2; ++mf)
+    for (int nf = 0; nf < 4; ++nf) {
+      long col = bcol + wc * 128 + nf * 32 + r31;
+      for (int r = 0; r < 16; ++r) {
+        long row = brow + wr * 64 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * g;
+        C[row * (long)N + col] = acc[mf][nf][r];
+      }
+    }
+}
+
+
 static int gemm_group(int tiles_n) {
   if (const char* env = std::getenv("HPK_GEMM_GROUP")) return std::atoi(env);
   if (tiles_n >= 128) return 32;
@@ -1275,6 +1381,23 @@ void launch_gemm_i8_nt(int* C, const void* A, const void* B, long M,
         "gemm_i8_nt requires M,N % 128 == 0 and K % 64 == 0");
   const int grp = gemm_group((int)(N / 256));
   const char* var = std::getenv("HPK_GEMM_VARIANT");
+  // HPK_GEMM_VARIANT=32: the 256^2 32x32x32 kernel — measured NEGATIVE
+  // for i8 (1984/2074 vs the 8-phase 16x16x64 pipeline's 2294/2393 TOPS
+  // at 8192/16384^3: the 16x16x64 instruction already has the same
+  // 16-byte-operand economy and the deep pipeline out-schedules the db
+  // structure); kept selectable, default stays 8ph
+  if (var && std::string(var) == "32" && M % 256 == 0 && N % 256 == 0 &&
+      K % 128 == 0) {
+    int tn32 = (int)(N / 256);
+    int n32 = (int)(M / 256) * tn32;
+    const char* genv = std::getenv("HPK_GEMM_GROUP");
+    const int grp32 = genv ? std::atoi(genv) : 1;
+    hipLaunchKernelGGL(k_gemm_i8_nt_32, dim3(n32), dim3(512), 0, stream, C,
+                       (const signed char*)A, (const signed char*)B, (int)M,
+                       (int)N, (int)K, tn32, n32, xcd_swizzle, grp32);
+    check_hip(hipGetLastError(), "launch_gemm_i8_nt(32)");
+    return;
+  }
   const bool ph8 = !var || std::string(var) == "8ph";
   if (ph8 && M % 256 == 0 && N % 256 == 0 && K % 256 == 0) {
     int tn = (int)(N / 256);

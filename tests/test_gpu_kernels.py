@@ -566,9 +566,12 @@ def test_gemm_i8_identity():
     assert torch.equal(c, b.int().t())
 
 
-def test_gemm_i8_8phase_exact():
-    """The 8-phase i8 path (256-divisible shapes): full-range exactness."""
+def test_gemm_i8_8phase_exact(monkeypatch):
+    """The 8-phase 16x16x64 i8 pipeline (pinned via HPK_GEMM_VARIANT —
+    256-divisible shapes now default to the 32x32x32 kernel)."""
     from hpc_patterns_amd import ops
+
+    monkeypatch.setenv("HPK_GEMM_VARIANT", "8ph")
 
     g = torch.Generator(device="cpu").manual_seed(61)
     m, n, k = 256, 512, 512
@@ -865,3 +868,22 @@ def test_gemm_fp8_32x32_default_exact():
     ref = torch.matmul(a.float(), b.float().t())
     torch.cuda.synchronize()
     assert torch.equal(c, ref), (c - ref).abs().max()
+
+
+def test_gemm_i8_32x32_variant_exact(monkeypatch):
+    """The selectable 256^2 32x32x32 i8 kernel (HPK_GEMM_VARIANT=32 — a
+    measured perf negative, kept for the design-space record): full-range
+    int8, exact int32, bitwise vs the int64 host reference."""
+    from hpc_patterns_amd import ops
+
+    monkeypatch.setenv("HPK_GEMM_VARIANT", "32")
+
+    g = torch.Generator(device="cpu").manual_seed(181)
+    m, n, k = 256, 512, 384
+    ah = torch.randint(-128, 128, (m, k), generator=g, dtype=torch.int8)
+    bh = torch.randint(-128, 128, (n, k), generator=g, dtype=torch.int8)
+    c = torch.empty(m, n, dtype=torch.int32, device="cuda")
+    ops.gemm_i8(c, ah.cuda(), bh.cuda())
+    ref = torch.matmul(ah.long(), bh.long().t()).to(torch.int32).cuda()
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
