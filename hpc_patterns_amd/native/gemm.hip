@@ -2098,6 +2098,155 @@ __global__ __launch_bounds__(512) void k_gemm_mxfp4_nt_32(
     }
 }
 
+
+// 256x128-tile variant of the 32x32x64 mx4 kernel: the occupancy
+// experiment the final PMC motivates (181 VGPR x 2 waves/SIMD = one
+// resident block; nothing covers the barrier windows, MfmaUtil 35%).
+// Halving the N-tile drops the accumulator floor to 64 VGPRs/lane so
+// TWO 8-wave blocks co-reside (4 waves/SIMD) and cover each other's
+// waits — at the price of arithmetic intensity (341 vs 508
+// FLOP/staged-byte). 8 waves as 4x2 of 64x64; per-wave glds counts
+// differ (A 2 issues + B 1 for all waves, + 1 scale issue for waves
+// 0-5), so the counted vmcnt branches per wave.
+__global__ __launch_bounds__(512) void k_gemm_mxfp4_nt_32h(
+    float* __restrict__ C, const unsigned char* __restrict__ A,
+    const unsigned char* __restrict__ B, const unsigned char* __restrict__ As,
+    const unsigned char* __restrict__ Bs, int M, int N, int K, int tiles_n,
+    int nwg, int xcd_swizzle, int group) {
+  constexpr int ATILE = 256 * 64; // 16 KiB
+  constexpr int BTILE = 128 * 64; // 8 KiB
+  __shared__ unsigned char lds[2 * (ATILE + BTILE) + 2 * 1536];
+  unsigned char* const sbase = lds + 2 * (ATILE + BTILE);
+
+  int wg = (int)blockIdx.x;
+  if (xcd_swizzle) {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = wg % 8, i = wg / 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  }
+  wg = hpk_group_remap(wg, tiles_n, nwg, group);
+  const long brow = (long)(wg / tiles_n) * 256;
+  const long bcol = (long)(wg % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1; // 4 wave-rows of 64
+  const int wc = wid & 1;  // 2 wave-cols of 64
+  const int ks = K / 32;
+  const long Kb = (long)K / 2;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  typedef __attribute__((ext_vector_type(4))) int i32x4;
+  f32x16 acc[2][2] = {};
+
+  auto stage = [&](int buf, int k0) {
+    unsigned char* dst = lds + (long)buf * (ATILE + BTILE);
+    for (int issue = 0; issue < 2; ++issue) { // A: 256 rows
+      long o_base = (long)issue * 8192 + (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 6);
+      int p = (int)((o & 63) >> 4);
+      int kk = ((p - (row >> 3)) & 3) * 16;
+      const unsigned char* ga = A + (brow + row) * Kb + k0 / 2 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)ga,
+          (__attribute__((address_space(3))) void*)(dst + o_base), 16, 0, 0);
+    }
+    { // B: 128 rows, one issue
+      long o_base = (long)wid * 1024;
+      long o = o_base + (long)lane * 16;
+      int row = (int)(o >> 6);
+      int p = (int)((o & 63) >> 4);
+      int kk = ((p - (row >> 3)) & 3) * 16;
+      const unsigned char* gb = B + (bcol + row) * Kb + k0 / 2 + kk;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gb,
+          (__attribute__((address_space(3))) void*)(dst + ATILE + o_base), 16,
+          0, 0);
+    }
+    if (wid < 6) { // scales: A rows 0-255 (waves 0-3), B rows 0-127 (4-5)
+      const unsigned char* S = (wid < 4) ? As : Bs;
+      long rbase = (wid < 4) ? brow : bcol;
+      int srow = (wid < 4 ? (wid & 3) : (wid & 1)) * 64 + lane;
+      const unsigned char* gs = S + (rbase + srow) * (long)ks + k0 / 32;
+      unsigned char* sdst = sbase + (long)buf * 1536 +
+                            (wid < 4 ? (wid & 3) * 256 : 1024 + (wid & 1) * 256);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gs,
+          (__attribute__((address_space(3))) void*)sdst, 4, 0, 0);
+    }
+  };
+
+  const int g = lane >> 5;
+  const int r31 = lane & 31;
+  int a_off[2], asc_off[2], b_off[2], bsc_off[2];
+  for (int mf = 0; mf < 2; ++mf) {
+    int row = wr * 64 + mf * 32 + r31;
+    a_off[mf] = row * 64;
+    asc_off[mf] = row * 4 + g;
+  }
+  for (int nf = 0; nf < 2; ++nf) {
+    int col = wc * 64 + nf * 32 + r31;
+    b_off[nf] = col * 64;
+    bsc_off[nf] = col * 4 + g;
+  }
+  i32x8 afrag[2] = {};
+  i32x8 bfrag[2] = {};
+
+  stage(0, 0);
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    const int cur = (k0 >> 7) & 1;
+    const bool more = (k0 + 128) < K;
+    if (more) stage(cur ^ 1, k0 + 128);
+    if (more) {
+      if (wid < 6)
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const unsigned char* la = lds + (long)cur * (ATILE + BTILE);
+    const unsigned char* lb = la + ATILE;
+    const unsigned char* sA = sbase + (long)cur * 1536;
+    const unsigned char* sB = sA + 1024;
+    for (int kk = 0; kk < 2; ++kk) {
+      int asc[2];
+      for (int mf = 0; mf < 2; ++mf) {
+        int row = wr * 64 + mf * 32 + r31;
+        int ch = (g + 2 * kk + (row >> 3)) & 3;
+        *(i32x4*)&afrag[mf] = *(const i32x4*)__builtin_assume_aligned(
+            la + a_off[mf] + 16 * ch, 16);
+        asc[mf] = sA[asc_off[mf] + 2 * kk];
+      }
+      for (int nf = 0; nf < 2; ++nf) {
+        int col = wc * 64 + nf * 32 + r31;
+        int ch = (g + 2 * kk + (col >> 3)) & 3;
+        *(i32x4*)&bfrag[nf] = *(const i32x4*)__builtin_assume_aligned(
+            lb + b_off[nf] + 16 * ch, 16);
+        int bsc = sB[bsc_off[nf] + 2 * kk];
+        for (int mf = 0; mf < 2; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              afrag[mf], bfrag[nf], acc[mf][nf], 4, 4, 0, asc[mf], 0, bsc);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int mf = 0; mf < 2; ++mf)
+    for (int nf = 0; nf < 2; ++nf) {
+      long col = bcol + wc * 64 + nf * 32 + r31;
+      for (int r = 0; r < 16; ++r) {
+        long row = brow + wr * 64 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * g;
+        C[row * (long)N + col] = acc[mf][nf][r];
+      }
+    }
+}
+
 void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
                           const void* As, const void* Bs, long M, long N,
                           long K, hipStream_t stream, int xcd_swizzle) {
@@ -2125,7 +2274,16 @@ void launch_gemm_mxfp4_nt(float* C, const void* A, const void* B,
                        (const unsigned char*)As, (const unsigned char*)Bs,
                        (int)M, (int)N, (int)K, tn32, n32, xcd_swizzle,
                        grp32);
-  } else if (waves == "db" || waves == "32") {
+  } else if (waves == "32h" && M % 256 == 0) {
+    int tnh = (int)(N / 128);
+    int nh = (int)(M / 256) * tnh;
+    const char* genv = std::getenv("HPK_GEMM_GROUP");
+    const int grph = genv ? std::atoi(genv) : 1;
+    hipLaunchKernelGGL(k_gemm_mxfp4_nt_32h, dim3(nh), dim3(512), 0, stream,
+                       C, (const unsigned char*)A, (const unsigned char*)B,
+                       (const unsigned char*)As, (const unsigned char*)Bs,
+                       (int)M, (int)N, (int)K, tnh, nh, xcd_swizzle, grph);
+  } else if (waves == "db" || waves == "32" || waves == "32h") {
     hipLaunchKernelGGL(k_gemm_mxfp4_nt_db, dim3(nwg), dim3(512), 0, stream,
                        C, (const unsigned char*)A, (const unsigned char*)B,
                        (const unsigned char*)As, (const unsigned char*)Bs,

@@ -887,3 +887,26 @@ def test_gemm_i8_32x32_variant_exact(monkeypatch):
     ref = torch.matmul(ah.long(), bh.long().t()).to(torch.int32).cuda()
     torch.cuda.synchronize()
     assert torch.equal(c, ref)
+
+
+def test_gemm_mxfp4_32h_variant_exact(monkeypatch):
+    """The 256x128 occupancy-experiment kernel (HPK_MX4_WAVES=32h, a
+    measured perf negative kept as the design-space record): exactness
+    with random power-of-2 scales."""
+    from hpc_patterns_amd import ops
+
+    monkeypatch.setenv("HPK_MX4_WAVES", "32h")
+    m, n, k = 512, 384, 256
+    fa, fb, pa, pb = _mx4_operands(m, n, k, 191)
+    g = torch.Generator(device="cpu").manual_seed(193)
+    sa = torch.randint(124, 131, (m, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    sb = torch.randint(124, 131, (n, k // 32), generator=g,
+                       dtype=torch.int16).to(torch.uint8)
+    da = fa * torch.pow(2.0, sa.float() - 127).repeat_interleave(32, dim=1)
+    db = fb * torch.pow(2.0, sb.float() - 127).repeat_interleave(32, dim=1)
+    ref = torch.matmul(da, db.t()).cuda()
+    c = torch.empty(m, n, dtype=torch.float32, device="cuda")
+    ops.gemm_mxfp4(c, pa, pb, sa.cuda(), sb.cuda())
+    torch.cuda.synchronize()
+    assert torch.equal(c, ref)
