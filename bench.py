@@ -189,6 +189,33 @@ def _measure_gemm_tf(step, size: int = 4096, iters: int = 3) -> float:
     return 2.0 * size ** 3 / best / 1e12
 
 
+def _measure_gemm_mx4_tf(step, size: int = 4096, iters: int = 3) -> float:
+    """K7-mx4 OCP MX-fp4 GEMM TFLOP/s at size^3 (random packed nibbles,
+    unit scales) — the family's fastest member (256^2 32x32x64 kernel)."""
+    import time
+
+    import torch
+
+    from hpc_patterns_amd import ops
+
+    p4a = torch.randint(0, 256, (size, size // 2), dtype=torch.uint8,
+                        device=step.device)
+    p4b = torch.randint(0, 256, (size, size // 2), dtype=torch.uint8,
+                        device=step.device)
+    s1 = torch.full((size, size // 32), 127, dtype=torch.uint8,
+                    device=step.device)
+    c = torch.empty(size, size, dtype=torch.float32, device=step.device)
+    ops.gemm_mxfp4(c, p4a, p4b, s1, s1)  # warmup
+    torch.cuda.synchronize()
+    best = float("inf")
+    for _ in range(iters):
+        t0 = time.perf_counter()
+        ops.gemm_mxfp4(c, p4a, p4b, s1, s1)
+        torch.cuda.synchronize()
+        best = min(best, time.perf_counter() - t0)
+    return 2.0 * size ** 3 / best / 1e12
+
+
 def _measure_mfma_tf(step, n_waves: int = 2048, tripcount: int = 20000,
                      iters: int = 3) -> float:
     """bf16 MFMA busy-loop TFLOP/s (v_mfma_f32_16x16x32_bf16 chains):
@@ -398,6 +425,8 @@ def main() -> int:
                    lambda: round(_measure_mfma_tf(step), 1))
         _component(components, "gemm_TFbf16",
                    lambda: round(_measure_gemm_tf(step), 1))
+        _component(components, "gemm_TFmx4",
+                   lambda: round(_measure_gemm_mx4_tf(step), 1))
     if world > 1:
         def _p2p():
             bw = pairwise_bandwidth(cfg["p2p_floats"] * 4, iters=5,
