@@ -19,21 +19,31 @@
 //                             at tile switches — DEFAULT for eligible
 //                             shapes: 968/1085 TF, zero LDS bank
 //                             conflicts (true-lane-group cyclic skew)
-//   k_gemm_fp8_nt / _8ph      OCP fp8 e4m3 twins (same MFMA rate, half
-//                             the staging bytes): up to 1264 TF
+//   k_gemm_fp8_nt / _8ph      OCP fp8 e4m3 at the 16x16x32 MFMA (half
+//                             the bf16 staging bytes): up to 1441 TF;
+//                             256-divisible shapes default to the
+//                             mxfp8_nt_32<false> route below (2.2 PF)
 //   k_gemm_mxfp8_nt           block-scaled OCP MX-fp8 via
 //                             mfma_scale_f32_16x16x128_f8f6f4 (HW-fused
 //                             e8m0 dequant; scale lane layout
 //                             reverse-engineered on hardware —
-//                             scripts/probes/): 1216 TF
+//                             scripts/probes/): 1216-1482 TF
+//   k_gemm_mxfp8_nt_32<S>     256^2-tile 32x32x64 version — DEFAULT for
+//                             256-divisible shapes: 1863 TF scaled (S=
+//                             true), 2194 TF as the plain-fp8 route
+//                             (S=false, hardcoded x1.0 scales); a THIRD
+//                             operand/scale association, half-interleaved
+//                             (findings #24)
 //   k_gemm_mxfp4_nt*          block-scaled OCP MX-fp4 (e2m1, the 4x rate
 //                             class): hardware-probed DIAGONAL layout;
 //                             default _32 kernel (256^2 tile, 32x32x64
 //                             MFMA, chunk-rotation bank fix):
 //                             2903/3044-3197 TF at 8192^3/16384^3
-//   k_gemm_i8_nt / _8ph       int8 with EXACT int32 accumulation
+//   k_gemm_i8_nt / _8ph / _32 int8 with EXACT int32 accumulation
 //                             (mfma_i32_16x16x64_i8, ~2x bf16 rate,
-//                             4-VGPR fragments): 1522 / 2214 TOPS
+//                             4-VGPR fragments): 1522 / 2214-2393 TOPS
+//                             (8ph stays default; the _32 port measured
+//                             negative, findings #25)
 //
 // Fragment mappings (16x16x32 bf16/fp8): A/B lane L holds 8 contiguous K
 // elements at k = 8*(L>>4), row/col = L&15; C/D lane L reg r holds
